@@ -1,0 +1,546 @@
+/*
+ * nvshare-scheduler — per-node GPU arbiter for nvshare-amd (MI355X).
+ *
+ * Serializes GPU work across transparent-sharing clients with an FCFS
+ * exclusive lock held for a time quantum (TQ).  Single-threaded epoll
+ * event loop plus one timer thread.  Wire-compatible with the reference
+ * protocol (see proto.h); behavioral parity reference:
+ * /root/reference/src/scheduler.c (registry, FCFS queue, TQ timer,
+ * strict eviction, SCHED_ON/OFF broadcast, SET_TQ).
+ *
+ * nvshare-amd additions over the reference:
+ *   - env-configurable startup state: NVSHARE_TQ (seconds),
+ *     NVSHARE_SCHED_OFF=1 (start with scheduling disabled)
+ *     (reference left these as TODOs, scheduler.c:549-552);
+ *   - solo-client fast path: the TQ timer does not preempt the lock
+ *     holder when nobody else is queued, so a lone client never pays
+ *     periodic drain stalls; a newly arriving waiter preempts the
+ *     holder as soon as the holder has had >= one full quantum;
+ *   - STATUS_REQ/STATUS query for observability (nvsharectl -q);
+ *   - partial-read tolerant framing (per-connection receive buffer);
+ *   - SIGUSR1 dumps scheduler state to stderr.
+ */
+#define _GNU_SOURCE
+#include <errno.h>
+#include <fcntl.h>
+#include <inttypes.h>
+#include <pthread.h>
+#include <signal.h>
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <sys/epoll.h>
+#include <sys/socket.h>
+#include <sys/stat.h>
+#include <unistd.h>
+
+#include "common.h"
+#include "proto.h"
+
+#define NVS_DEFAULT_TQ 30
+#define MAX_EPOLL_EVENTS 64
+
+/* One TCP^WUnix connection. Becomes a "client" after REGISTER. */
+struct conn {
+	int fd;
+	size_t got;               /* bytes of in-progress message */
+	struct nvs_msg inmsg;
+	int registered;
+	uint64_t id;              /* client id (registered only) */
+	char pod_name[NVS_POD_NAME_LEN];
+	char pod_namespace[NVS_POD_NS_LEN];
+	int wants_lock;           /* present in request queue */
+	struct conn *next;        /* registry list */
+	struct conn *qnext;       /* FCFS queue list */
+};
+
+static struct conn *clients;      /* all connections (incl. unregistered) */
+static struct conn *queue_head;   /* FCFS lock queue */
+static struct conn *queue_tail;
+
+static pthread_mutex_t g_mutex = PTHREAD_MUTEX_INITIALIZER;
+static pthread_cond_t timer_cv = PTHREAD_COND_INITIALIZER;
+
+static int scheduler_on = 1;
+static int tq_seconds = NVS_DEFAULT_TQ;
+static int lock_held;
+static struct conn *lock_holder;
+static int drop_lock_sent;
+static unsigned long scheduling_round;
+static int64_t quantum_start_ns;
+/* stats */
+static unsigned long total_grants, total_preemptions, total_evictions;
+
+static int epoll_fd = -1;
+static char sock_path[NVS_SOCK_PATH_MAX];
+static volatile sig_atomic_t dump_requested;
+
+static int queue_len(void)
+{
+	int n = 0;
+	struct conn *c;
+
+	for (c = queue_head; c != NULL; c = c->qnext)
+		n++;
+	return n;
+}
+
+static int client_count(void)
+{
+	int n = 0;
+	struct conn *c;
+
+	for (c = clients; c != NULL; c = c->next)
+		if (c->registered)
+			n++;
+	return n;
+}
+
+static void send_to(struct conn *c, uint8_t type, const char *data)
+{
+	struct nvs_msg m;
+
+	nvs_msg_init(&m, type, c->id, data);
+	if (nvs_send_msg(c->fd, &m) != 0)
+		log_warn("send %s to client %016" PRIx64 " failed",
+			 nvs_msg_type_str(type), c->id);
+	else
+		log_debug("sent %s to client %016" PRIx64,
+			  nvs_msg_type_str(type), c->id);
+}
+
+static void queue_push(struct conn *c)
+{
+	c->qnext = NULL;
+	if (queue_tail != NULL)
+		queue_tail->qnext = c;
+	else
+		queue_head = c;
+	queue_tail = c;
+	c->wants_lock = 1;
+}
+
+static void queue_remove(struct conn *c)
+{
+	struct conn **pp = &queue_head;
+	struct conn *prev = NULL, *it;
+
+	for (it = queue_head; it != NULL; prev = it, it = it->qnext) {
+		if (it == c) {
+			*pp = it->qnext;
+			if (queue_tail == c)
+				queue_tail = prev;
+			break;
+		}
+		pp = &it->qnext;
+	}
+	c->qnext = NULL;
+	c->wants_lock = 0;
+}
+
+/* Grant the lock to the queue head if possible. Caller holds g_mutex. */
+static void try_schedule(void)
+{
+	if (!scheduler_on || lock_held || queue_head == NULL)
+		return;
+	lock_holder = queue_head;
+	lock_held = 1;
+	drop_lock_sent = 0;
+	scheduling_round++;
+	total_grants++;
+	quantum_start_ns = nvs_now_ns();
+	send_to(lock_holder, NVS_LOCK_OK, NULL);
+	log_debug("round %lu: lock -> %016" PRIx64 " (queue=%d)",
+		  scheduling_round, lock_holder->id, queue_len());
+	pthread_cond_broadcast(&timer_cv);
+}
+
+/*
+ * TQ timer thread.  While a client holds the lock and at least one other
+ * client is queued, send DROP_LOCK once the holder has had tq seconds.
+ * A lone holder is never preempted (solo fast path); the arrival of a
+ * waiter wakes this thread, which then preempts as soon as the holder's
+ * cumulative quantum is exhausted.  After DROP_LOCK we wait indefinitely
+ * for LOCK_RELEASED; only socket death evicts a stuck client (strict
+ * eviction, as the reference: scheduler.c:352,645-663).
+ */
+static void *timer_thread(void *arg)
+{
+	(void)arg;
+	pthread_mutex_lock(&g_mutex);
+	for (;;) {
+		while (!lock_held || !scheduler_on || drop_lock_sent ||
+		       queue_head == NULL || queue_head->qnext == NULL) {
+			pthread_cond_wait(&timer_cv, &g_mutex);
+		}
+		/* Holder + >=1 waiter: arm the deadline. */
+		{
+			unsigned long round = scheduling_round;
+			int64_t deadline_ns =
+				quantum_start_ns +
+				(int64_t)tq_seconds * 1000000000LL;
+			int64_t now = nvs_now_ns();
+
+			if (now < deadline_ns) {
+				struct timespec abs;
+				clock_gettime(CLOCK_REALTIME, &abs);
+				int64_t wait_ns = deadline_ns - now;
+				abs.tv_sec += wait_ns / 1000000000LL;
+				abs.tv_nsec += wait_ns % 1000000000LL;
+				if (abs.tv_nsec >= 1000000000L) {
+					abs.tv_sec++;
+					abs.tv_nsec -= 1000000000L;
+				}
+				pthread_cond_timedwait(&timer_cv, &g_mutex,
+						       &abs);
+				/* State may have changed; re-evaluate. */
+				if (scheduling_round != round)
+					continue;
+				if (nvs_now_ns() <
+				    quantum_start_ns +
+				    (int64_t)tq_seconds * 1000000000LL)
+					continue;
+			}
+			if (scheduling_round != round || !lock_held ||
+			    !scheduler_on || drop_lock_sent ||
+			    queue_head == NULL || queue_head->qnext == NULL)
+				continue;
+			drop_lock_sent = 1;
+			total_preemptions++;
+			log_debug("round %lu: TQ expired, DROP_LOCK -> "
+				  "%016" PRIx64, scheduling_round,
+				  lock_holder->id);
+			send_to(lock_holder, NVS_DROP_LOCK, NULL);
+		}
+	}
+	return NULL;
+}
+
+static void bcast_status(void)
+{
+	struct conn *c;
+
+	for (c = clients; c != NULL; c = c->next)
+		if (c->registered)
+			send_to(c, scheduler_on ? NVS_SCHED_ON :
+						  NVS_SCHED_OFF, NULL);
+}
+
+static void delete_conn(struct conn *c)
+{
+	struct conn **pp;
+
+	epoll_ctl(epoll_fd, EPOLL_CTL_DEL, c->fd, NULL);
+	close(c->fd);
+
+	if (c->registered) {
+		total_evictions++;
+		log_info("client %016" PRIx64 " (%s/%s) disconnected",
+			 c->id, c->pod_namespace[0] ? c->pod_namespace : "-",
+			 c->pod_name[0] ? c->pod_name : "-");
+	}
+	if (c->wants_lock)
+		queue_remove(c);
+	if (lock_holder == c) {
+		lock_held = 0;
+		lock_holder = NULL;
+		drop_lock_sent = 0;
+	}
+	for (pp = &clients; *pp != NULL; pp = &(*pp)->next) {
+		if (*pp == c) {
+			*pp = c->next;
+			break;
+		}
+	}
+	free(c);
+	try_schedule();
+}
+
+static void handle_status_req(struct conn *c)
+{
+	struct nvs_msg m;
+	char buf[NVS_MSG_DATA_LEN];
+
+	/* data: "<on>,<tq>,<nclients>,<qlen>" — fits in 20 bytes for sane
+	 * values; counts are clamped by the field width. */
+	snprintf(buf, sizeof(buf), "%d,%d,%d,%d", scheduler_on, tq_seconds,
+		 client_count(), queue_len());
+	nvs_msg_init(&m, NVS_STATUS, c->id, buf);
+	if (nvs_send_msg(c->fd, &m) != 0)
+		log_warn("STATUS reply failed");
+}
+
+static void process_msg(struct conn *c, const struct nvs_msg *m)
+{
+	log_debug("recv %s from fd=%d id=%016" PRIx64,
+		  nvs_msg_type_str(m->type), c->fd, c->id);
+
+	switch (m->type) {
+	case NVS_REGISTER: {
+		char idbuf[NVS_MSG_DATA_LEN];
+
+		if (c->registered) {
+			log_warn("duplicate REGISTER from %016" PRIx64, c->id);
+			break;
+		}
+		c->registered = 1;
+		c->id = nvs_gen_id();
+		memcpy(c->pod_name, m->pod_name, NVS_POD_NAME_LEN);
+		c->pod_name[NVS_POD_NAME_LEN - 1] = '\0';
+		memcpy(c->pod_namespace, m->pod_namespace, NVS_POD_NS_LEN);
+		c->pod_namespace[NVS_POD_NS_LEN - 1] = '\0';
+		snprintf(idbuf, sizeof(idbuf), "%016" PRIx64, c->id);
+		log_info("registered client %016" PRIx64 " (%s/%s)", c->id,
+			 c->pod_namespace[0] ? c->pod_namespace : "-",
+			 c->pod_name[0] ? c->pod_name : "-");
+		send_to(c, scheduler_on ? NVS_SCHED_ON : NVS_SCHED_OFF,
+			idbuf);
+		break;
+	}
+	case NVS_REQ_LOCK:
+		if (!c->registered) {
+			log_warn("REQ_LOCK from unregistered fd=%d", c->fd);
+			break;
+		}
+		if (!scheduler_on) {
+			/* Clients free-run while scheduling is off. */
+			break;
+		}
+		if (c->wants_lock || lock_holder == c)
+			break;
+		queue_push(c);
+		if (!lock_held)
+			try_schedule();
+		else
+			pthread_cond_broadcast(&timer_cv); /* arm preemption */
+		break;
+	case NVS_LOCK_RELEASED:
+		if (lock_holder != c) {
+			/* Late/duplicate release (e.g. early-release racing
+			 * a preemption) — ignore. */
+			log_debug("stale LOCK_RELEASED from %016" PRIx64,
+				  c->id);
+			break;
+		}
+		queue_remove(c);
+		lock_held = 0;
+		lock_holder = NULL;
+		drop_lock_sent = 0;
+		try_schedule();
+		break;
+	case NVS_SET_TQ: {
+		char buf[NVS_MSG_DATA_LEN];
+		long v;
+		char *end = NULL;
+
+		memcpy(buf, m->data, NVS_MSG_DATA_LEN);
+		buf[NVS_MSG_DATA_LEN - 1] = '\0';
+		v = strtol(buf, &end, 10);
+		if (end == buf || v < 1 || v > 86400) {
+			log_warn("SET_TQ: invalid value '%s'", buf);
+			break;
+		}
+		tq_seconds = (int)v;
+		log_info("TQ set to %d s", tq_seconds);
+		pthread_cond_broadcast(&timer_cv);
+		break;
+	}
+	case NVS_SCHED_ON:
+		if (!scheduler_on) {
+			scheduler_on = 1;
+			log_info("scheduling enabled");
+			bcast_status();
+			try_schedule();
+		}
+		break;
+	case NVS_SCHED_OFF:
+		if (scheduler_on) {
+			struct conn *it, *nx;
+
+			scheduler_on = 0;
+			log_info("scheduling disabled (free-for-all)");
+			bcast_status();
+			/* Flush the queue; everyone may run. */
+			for (it = queue_head; it != NULL; it = nx) {
+				nx = it->qnext;
+				it->qnext = NULL;
+				it->wants_lock = 0;
+			}
+			queue_head = queue_tail = NULL;
+			lock_held = 0;
+			lock_holder = NULL;
+			drop_lock_sent = 0;
+		}
+		break;
+	case NVS_STATUS_REQ:
+		handle_status_req(c);
+		break;
+	default:
+		log_warn("unknown message type %u from fd=%d", m->type,
+			 c->fd);
+		break;
+	}
+}
+
+/* Drain readable bytes from a connection; returns -1 when it died. */
+static int handle_readable(struct conn *c)
+{
+	for (;;) {
+		char *base = (char *)&c->inmsg;
+		ssize_t n = read(c->fd, base + c->got,
+				 NVS_MSG_SIZE - c->got);
+
+		if (n == 0)
+			return -1;
+		if (n < 0) {
+			if (errno == EINTR)
+				continue;
+			if (errno == EAGAIN || errno == EWOULDBLOCK)
+				return 0;
+			return -1;
+		}
+		c->got += (size_t)n;
+		if (c->got == NVS_MSG_SIZE) {
+			c->got = 0;
+			process_msg(c, &c->inmsg);
+		}
+	}
+}
+
+static void dump_state(void)
+{
+	struct conn *c;
+
+	log_info("=== scheduler state ===");
+	log_info("on=%d tq=%ds lock_held=%d holder=%016" PRIx64
+		 " round=%lu grants=%lu preempts=%lu evictions=%lu",
+		 scheduler_on, tq_seconds, lock_held,
+		 lock_holder ? lock_holder->id : 0, scheduling_round,
+		 total_grants, total_preemptions, total_evictions);
+	for (c = clients; c != NULL; c = c->next)
+		log_info("  conn fd=%d reg=%d id=%016" PRIx64 " queued=%d%s",
+			 c->fd, c->registered, c->id, c->wants_lock,
+			 lock_holder == c ? " [HOLDER]" : "");
+}
+
+static void on_sigusr1(int sig)
+{
+	(void)sig;
+	dump_requested = 1;
+}
+
+int main(void)
+{
+	int lsock;
+	pthread_t timer_tid;
+	struct epoll_event ev, events[MAX_EPOLL_EVENTS];
+	const char *dir;
+	char dirbuf[NVS_SOCK_PATH_MAX];
+	struct sigaction sa;
+
+	nvs_log_init();
+	signal(SIGPIPE, SIG_IGN);
+	memset(&sa, 0, sizeof(sa));
+	sa.sa_handler = on_sigusr1;
+	sigaction(SIGUSR1, &sa, NULL);
+
+	tq_seconds = (int)nvs_env_long("NVSHARE_TQ", NVS_DEFAULT_TQ, 1,
+				       86400);
+	scheduler_on = !nvs_env_bool("NVSHARE_SCHED_OFF", 0);
+
+	/* Ensure the socket directory exists. */
+	dir = getenv(NVS_SOCK_DIR_ENV);
+	if (dir == NULL || dir[0] == '\0')
+		dir = NVS_SOCK_DIR_DEFAULT;
+	nvs_strlcpy(dirbuf, dir, sizeof(dirbuf));
+	mkdir(dirbuf, 0777);
+
+	true_or_exit(nvs_scheduler_path(sock_path) == 0);
+	lsock = nvs_bind_listen(sock_path);
+	if (lsock < 0)
+		log_fatal("cannot bind %s: %s", sock_path, strerror(errno));
+
+	epoll_fd = epoll_create1(EPOLL_CLOEXEC);
+	true_or_exit(epoll_fd >= 0);
+	memset(&ev, 0, sizeof(ev));
+	ev.events = EPOLLIN;
+	ev.data.ptr = NULL; /* NULL => listen socket */
+	true_or_exit(epoll_ctl(epoll_fd, EPOLL_CTL_ADD, lsock, &ev) == 0);
+
+	true_or_exit(pthread_create(&timer_tid, NULL, timer_thread, NULL)
+		     == 0);
+
+	log_info("nvshare-scheduler (amd) listening on %s (tq=%ds, "
+		 "scheduling %s)", sock_path, tq_seconds,
+		 scheduler_on ? "on" : "off");
+
+	for (;;) {
+		int nev, i;
+
+		nev = epoll_wait(epoll_fd, events, MAX_EPOLL_EVENTS, 1000);
+		if (nev < 0) {
+			if (errno == EINTR) {
+				if (dump_requested) {
+					dump_requested = 0;
+					pthread_mutex_lock(&g_mutex);
+					dump_state();
+					pthread_mutex_unlock(&g_mutex);
+				}
+				continue;
+			}
+			log_fatal("epoll_wait: %s", strerror(errno));
+		}
+		if (dump_requested) {
+			dump_requested = 0;
+			pthread_mutex_lock(&g_mutex);
+			dump_state();
+			pthread_mutex_unlock(&g_mutex);
+		}
+		pthread_mutex_lock(&g_mutex);
+		for (i = 0; i < nev; i++) {
+			struct conn *c = events[i].data.ptr;
+
+			if (c == NULL) {
+				/* New connection(s). */
+				for (;;) {
+					int fd = nvs_accept(lsock);
+					struct conn *nc;
+
+					if (fd < 0)
+						break;
+					/* Nonblocking for partial reads. */
+					{
+						int fl = fcntl(fd, F_GETFL);
+						fcntl(fd, F_SETFL,
+						      fl | O_NONBLOCK);
+					}
+					nc = calloc(1, sizeof(*nc));
+					true_or_exit(nc != NULL);
+					nc->fd = fd;
+					nc->next = clients;
+					clients = nc;
+					memset(&ev, 0, sizeof(ev));
+					ev.events = EPOLLIN | EPOLLRDHUP;
+					ev.data.ptr = nc;
+					true_or_exit(epoll_ctl(epoll_fd,
+						EPOLL_CTL_ADD, fd, &ev) == 0);
+					log_debug("accepted fd=%d", fd);
+				}
+				continue;
+			}
+			if (events[i].events & (EPOLLERR | EPOLLHUP |
+						EPOLLRDHUP)) {
+				/* Drain anything pending, then evict. */
+				handle_readable(c);
+				delete_conn(c);
+				continue;
+			}
+			if (events[i].events & EPOLLIN) {
+				if (handle_readable(c) < 0)
+					delete_conn(c);
+			}
+		}
+		pthread_mutex_unlock(&g_mutex);
+	}
+	return 0;
+}
