@@ -1,0 +1,90 @@
+"""ResNet training job on synthetic data (co-location benchmark unit).
+
+One "step" = forward + loss + backward + SGD update on a synthetic
+batch.  Emits steps/s and samples/s.  This is the flagship workload of
+bench.py (BASELINE.json config #4: co-located ResNet-50 training).
+"""
+
+from __future__ import annotations
+
+import argparse
+
+from nvshare_amd.workloads.common import Timer, add_common_args, emit, sync
+
+
+def build(model_name: str, num_classes: int):
+    from nvshare_amd.workloads import resnet
+
+    if model_name == "resnet50":
+        return resnet.resnet50(num_classes)
+    if model_name == "resnet152":
+        return resnet.resnet152(num_classes)
+    if model_name == "tiny":
+        return resnet.tiny_resnet(num_classes)
+    raise ValueError(f"unknown model {model_name}")
+
+
+def run_training(model_name: str = "resnet50", device: str = "cuda",
+                 batch: int = 64, image: int = 224, steps: int = 50,
+                 warmup: int = 5, dtype: str = "bfloat16",
+                 num_classes: int = 1000) -> dict:
+    import torch
+
+    dev = torch.device(device)
+    amp_dtype = getattr(torch, dtype) if dtype != "float32" else None
+    torch.manual_seed(0)
+    model = build(model_name, num_classes).to(dev)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    lossf = torch.nn.CrossEntropyLoss()
+    x = torch.randn(batch, 3, image, image, device=dev)
+    y = torch.randint(0, num_classes, (batch,), device=dev)
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        if amp_dtype is not None and device.startswith("cuda"):
+            with torch.autocast("cuda", dtype=amp_dtype):
+                loss = lossf(model(x), y)
+        else:
+            loss = lossf(model(x), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    for _ in range(warmup):
+        step()
+    sync(device)
+    with Timer() as t:
+        for _ in range(steps):
+            loss = step()
+        sync(device)
+    final_loss = float(loss.detach().float().cpu())
+    return {
+        "workload": "train_resnet", "model": model_name,
+        "seconds": t.seconds, "steps": steps, "batch": batch,
+        "image": image, "device": device, "dtype": dtype,
+        "steps_per_s": steps / t.seconds,
+        "samples_per_s": steps * batch / t.seconds,
+        "loss": final_loss,
+    }
+
+
+def main(argv: list[str] | None = None) -> None:
+    ap = argparse.ArgumentParser()
+    add_common_args(ap)
+    ap.add_argument("--model", default="resnet50")
+    ap.add_argument("--batch", type=int, default=64)
+    ap.add_argument("--image", type=int, default=224)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--dtype", default="bfloat16")
+    ap.add_argument("--num-classes", type=int, default=1000)
+    args = ap.parse_args(argv)
+    res = run_training(args.model, args.device, args.batch, args.image,
+                       args.steps, args.warmup, args.dtype,
+                       args.num_classes)
+    res["label"] = args.label
+    emit(res)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,192 @@
+"""Interposer (libnvshare.so) tests against the CPU stub HIP runtime.
+
+Exercises the full LD_PRELOAD stack without a GPU: managed-memory
+conversion, allocation cap/oversubscription, hipMemGetInfo accounting,
+and scheduler serialization of two co-located clients (the thesis
+validated the CUDA original the same way, by counting interposed
+launches).
+"""
+
+from __future__ import annotations
+
+import os
+import subprocess
+import time
+from collections import defaultdict
+
+import pytest
+
+from nvshare_amd.env import client_env
+
+
+def run_hipclient(artifacts, sock_dir, *args, env_extra=None,
+                  stub_env=None, timeout=60, **env_kwargs):
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     **env_kwargs)
+    env.update(stub_env or {})
+    env.update(env_extra or {})
+    return subprocess.run(
+        [str(artifacts.hipclient), *map(str, args)],
+        env=env, capture_output=True, text=True, timeout=timeout,
+    )
+
+
+def load_events(path):
+    events = []
+    if not os.path.exists(path):
+        return events
+    with open(path) as f:
+        for line in f:
+            ts, pid, name, arg = line.split()
+            events.append((int(ts), int(pid), name, int(arg)))
+    return events
+
+
+def test_malloc_becomes_managed(artifacts, sched, sock_dir):
+    log = os.path.join(sock_dir, "ev.log")
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 2, "--alloc-mib",
+                      16, "--iters", 5,
+                      stub_env={"NVSTUB_LOG": log,
+                                "NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=64)
+    assert r.returncode == 0, r.stderr
+    assert "PASS" in r.stdout
+    names = [e[2] for e in load_events(log)]
+    assert "hipMallocManaged" in names
+    assert "hipMalloc" not in names  # every alloc was converted
+
+
+def test_allocation_cap_oom(artifacts, sched, sock_dir):
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      2000, "--iters", 1,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=64)
+    assert r.returncode == 3, (r.stdout, r.stderr)
+    assert "OOM" in r.stdout
+
+
+def test_oversubscription_env(artifacts, sched, sock_dir):
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      2000, "--iters", 1,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=64, oversubscribe=True)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "PASS" in r.stdout
+
+
+def test_fake_total_override(artifacts, sched, sock_dir):
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      300, "--iters", 1,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=64, fake_total_mib=256)
+    assert r.returncode == 3, (r.stdout, r.stderr)
+
+
+def test_memgetinfo_reports_reserve(artifacts, sched, sock_dir):
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      100, "--iters", 1,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=256)
+    assert r.returncode == 0
+    # hipclient prints: free=<x> MiB total=<y> MiB (after its alloc)
+    line = [l for l in r.stderr.splitlines() if "free=" in l][0]
+    free = int(line.split("free=")[1].split()[0])
+    total = int(line.split("total=")[1].split()[0])
+    assert total == 1024
+    assert free == 1024 - 256 - 100
+
+
+def test_standalone_no_scheduler(artifacts, sock_dir):
+    """NVSHARE_STANDALONE runs without any scheduler."""
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      8, "--iters", 3,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      reserve_mib=64, standalone=True)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+
+
+def test_connect_timeout_fatal(artifacts, sock_dir):
+    """Without a scheduler and without standalone, clients fail fast
+    with a clear error (the reference hung forever)."""
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      8, "--iters", 1,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      env_extra={"NVSHARE_CONNECT_TIMEOUT_S": "1"},
+                      reserve_mib=64, timeout=30)
+    assert r.returncode != 0
+    assert "cannot reach nvshare-scheduler" in r.stderr
+
+
+def test_two_clients_serialized(artifacts, sched, sock_dir):
+    """Kernel windows of co-located clients must never overlap."""
+    log = os.path.join(sock_dir, "ev2.log")
+    stub_env = {"NVSTUB_LOG": log, "NVSTUB_TOTAL_MIB": "1024",
+                "NVSTUB_KERNEL_US": "2500"}
+    env = client_env(sock_dir=sock_dir, use_stub=True, reserve_mib=64)
+    env.update(stub_env)
+    procs = []
+    for i in range(2):
+        e = dict(env)
+        e["NVSHARE_POD_NAME"] = f"cl{i}"
+        procs.append(subprocess.Popen(
+            [str(artifacts.hipclient), "--allocs", "1", "--alloc-mib",
+             "16", "--iters", "800"],
+            env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+            text=True))
+    for p in procs:
+        out, err = p.communicate(timeout=120)
+        assert p.returncode == 0, (out, err)
+        assert "PASS" in out
+
+    events = load_events(log)
+    intervals = []
+    open_t = {}
+    for ts, pid, name, arg in events:
+        if name == "launch_begin":
+            open_t[pid] = ts
+        elif name == "launch_end":
+            intervals.append((open_t[pid], ts, pid))
+    assert len(intervals) == 1600
+    intervals.sort()
+    pids = {iv[2] for iv in intervals}
+    assert len(pids) == 2
+    overlaps = sum(
+        1 for a, b in zip(intervals, intervals[1:])
+        if b[0] < a[1] and a[2] != b[2]
+    )
+    assert overlaps == 0
+
+    # The TQ timer (1s) must have rotated the lock between the clients.
+    sched_log = sched.log_text()
+    assert "DROP_LOCK" in sched_log
+
+
+def test_scheduler_restart_reconnect(artifacts, sock_dir):
+    """Clients survive a scheduler restart (reference killed the app)."""
+    from nvshare_amd.scheduler import SchedulerDaemon
+
+    log = os.path.join(sock_dir, "ev3.log")
+    stub_env = {"NVSTUB_LOG": log, "NVSTUB_TOTAL_MIB": "1024",
+                "NVSTUB_KERNEL_US": "20000"}
+    d1 = SchedulerDaemon(sock_dir=sock_dir, tq=1)
+    d1.start()
+    env = client_env(sock_dir=sock_dir, use_stub=True, reserve_mib=64)
+    env.update(stub_env)
+    env["NVSHARE_RECONNECT_S"] = "30"
+    p = subprocess.Popen(
+        [str(artifacts.hipclient), "--allocs", "1", "--alloc-mib", "8",
+         "--iters", "300"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+        text=True)
+    time.sleep(2)  # client is mid-run
+    d1.stop()
+    time.sleep(1)
+    d2 = SchedulerDaemon(sock_dir=sock_dir, tq=1)
+    d2.start()
+    try:
+        out, err = p.communicate(timeout=60)
+        assert p.returncode == 0, (out, err)
+        assert "PASS" in out
+        assert "reconnecting" in err
+    finally:
+        d2.stop()
