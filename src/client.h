@@ -46,6 +46,13 @@ struct nvs_real_hip {
 	fn_hipMemcpyDtoDAsync hipMemcpyDtoDAsync;
 	fn_hipMemset hipMemset;
 	fn_hipMemsetAsync hipMemsetAsync;
+	fn_hipMemsetD32Async hipMemsetD32Async;
+	fn_hipMemcpy2D hipMemcpy2D;
+	fn_hipMemcpy2DAsync hipMemcpy2DAsync;
+	fn_hipMemcpyToSymbol hipMemcpyToSymbol;
+	fn_hipMemcpyFromSymbol hipMemcpyFromSymbol;
+	fn_hipMemcpyPeerAsync hipMemcpyPeerAsync;
+	fn_hipModuleLaunchCooperativeKernel hipModuleLaunchCooperativeKernel;
 	fn_hipGetProcAddress hipGetProcAddress;
 	fn_hipGetErrorString hipGetErrorString;
 };

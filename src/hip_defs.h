@@ -112,6 +112,28 @@ typedef nvshipError_t (*fn_hipMemcpyDtoDAsync)(nvship_deviceptr_t,
 typedef nvshipError_t (*fn_hipMemset)(void *, int, size_t);
 typedef nvshipError_t (*fn_hipMemsetAsync)(void *, int, size_t,
 					   nvship_stream_t);
+typedef nvshipError_t (*fn_hipMemsetD32Async)(nvship_deviceptr_t, int,
+					      size_t, nvship_stream_t);
+typedef nvshipError_t (*fn_hipMemcpy2D)(void *, size_t, const void *,
+					size_t, size_t, size_t,
+					nvship_memcpy_kind);
+typedef nvshipError_t (*fn_hipMemcpy2DAsync)(void *, size_t, const void *,
+					     size_t, size_t, size_t,
+					     nvship_memcpy_kind,
+					     nvship_stream_t);
+typedef nvshipError_t (*fn_hipMemcpyToSymbol)(const void *, const void *,
+					      size_t, size_t,
+					      nvship_memcpy_kind);
+typedef nvshipError_t (*fn_hipMemcpyFromSymbol)(void *, const void *,
+						size_t, size_t,
+						nvship_memcpy_kind);
+typedef nvshipError_t (*fn_hipMemcpyPeerAsync)(void *, int, const void *,
+					       int, size_t,
+					       nvship_stream_t);
+typedef nvshipError_t (*fn_hipModuleLaunchCooperativeKernel)(
+	nvship_function_t, unsigned int, unsigned int, unsigned int,
+	unsigned int, unsigned int, unsigned int, unsigned int,
+	nvship_stream_t, void **);
 typedef nvshipError_t (*fn_hipGetProcAddress)(const char *, void **, int,
 					      uint64_t, void *);
 typedef const char *(*fn_hipGetErrorString)(nvshipError_t);

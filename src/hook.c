@@ -169,6 +169,20 @@ static void bootstrap(void)
 		(fn_hipMemcpyDtoDAsync)resolve("hipMemcpyDtoDAsync");
 	real.hipMemset = (fn_hipMemset)resolve("hipMemset");
 	real.hipMemsetAsync = (fn_hipMemsetAsync)resolve("hipMemsetAsync");
+	real.hipMemsetD32Async =
+		(fn_hipMemsetD32Async)resolve("hipMemsetD32Async");
+	real.hipMemcpy2D = (fn_hipMemcpy2D)resolve("hipMemcpy2D");
+	real.hipMemcpy2DAsync =
+		(fn_hipMemcpy2DAsync)resolve("hipMemcpy2DAsync");
+	real.hipMemcpyToSymbol =
+		(fn_hipMemcpyToSymbol)resolve("hipMemcpyToSymbol");
+	real.hipMemcpyFromSymbol =
+		(fn_hipMemcpyFromSymbol)resolve("hipMemcpyFromSymbol");
+	real.hipMemcpyPeerAsync =
+		(fn_hipMemcpyPeerAsync)resolve("hipMemcpyPeerAsync");
+	real.hipModuleLaunchCooperativeKernel =
+		(fn_hipModuleLaunchCooperativeKernel)
+		resolve("hipModuleLaunchCooperativeKernel");
 	real.hipGetProcAddress =
 		(fn_hipGetProcAddress)resolve("hipGetProcAddress");
 	real.hipGetErrorString =
@@ -358,6 +372,73 @@ static void after_launch(void)
 }
 
 /* ------------------------------------------------------------------ */
+/* Hook call counters (rocprofv3 cross-check of interposition          */
+/* completeness, the thesis Table 11.6 methodology: our per-symbol     */
+/* counts must equal the profiler's API counts).  Dumped at exit when  */
+/* NVSHARE_DEBUG is set.                                               */
+/* ------------------------------------------------------------------ */
+
+enum hook_id {
+	H_hipMalloc, H_hipMallocManaged, H_hipExtMallocWithFlags,
+	H_hipMallocAsync, H_hipMallocFromPoolAsync, H_hipFree,
+	H_hipFreeAsync, H_hipMemGetInfo, H_hipSetDevice,
+	H_hipLaunchKernel, H_hipExtLaunchKernel,
+	H_hipLaunchCooperativeKernel, H_hipModuleLaunchKernel,
+	H_hipExtModuleLaunchKernel, H_hipModuleLaunchCooperativeKernel,
+	H_hipGraphLaunch, H_hipMemcpy, H_hipMemcpyAsync,
+	H_hipMemcpyWithStream, H_hipMemcpyHtoD, H_hipMemcpyDtoH,
+	H_hipMemcpyDtoD, H_hipMemcpyHtoDAsync, H_hipMemcpyDtoHAsync,
+	H_hipMemcpyDtoDAsync, H_hipMemcpy2D, H_hipMemcpy2DAsync,
+	H_hipMemcpyToSymbol, H_hipMemcpyFromSymbol, H_hipMemcpyPeerAsync,
+	H_hipMemset, H_hipMemsetAsync, H_hipMemsetD32Async,
+	H_hipGetProcAddress, H_dlsym,
+	H_COUNT_
+};
+
+static const char *hook_names[H_COUNT_] = {
+	"hipMalloc", "hipMallocManaged", "hipExtMallocWithFlags",
+	"hipMallocAsync", "hipMallocFromPoolAsync", "hipFree",
+	"hipFreeAsync", "hipMemGetInfo", "hipSetDevice",
+	"hipLaunchKernel", "hipExtLaunchKernel",
+	"hipLaunchCooperativeKernel", "hipModuleLaunchKernel",
+	"hipExtModuleLaunchKernel", "hipModuleLaunchCooperativeKernel",
+	"hipGraphLaunch", "hipMemcpy", "hipMemcpyAsync",
+	"hipMemcpyWithStream", "hipMemcpyHtoD", "hipMemcpyDtoH",
+	"hipMemcpyDtoD", "hipMemcpyHtoDAsync", "hipMemcpyDtoHAsync",
+	"hipMemcpyDtoDAsync", "hipMemcpy2D", "hipMemcpy2DAsync",
+	"hipMemcpyToSymbol", "hipMemcpyFromSymbol", "hipMemcpyPeerAsync",
+	"hipMemset", "hipMemsetAsync", "hipMemsetD32Async",
+	"hipGetProcAddress", "dlsym",
+};
+
+static unsigned long hook_counts[H_COUNT_];
+
+#define BUMP(id) __atomic_fetch_add(&hook_counts[id], 1UL, \
+				    __ATOMIC_RELAXED)
+
+__attribute__((destructor)) static void dump_hook_counts(void)
+{
+	int i, any = 0;
+	char line[2048];
+	size_t off = 0;
+
+	if (!nvs_debug_enabled)
+		return;
+	for (i = 0; i < H_COUNT_; i++) {
+		if (hook_counts[i] == 0)
+			continue;
+		any = 1;
+		off += (size_t)snprintf(line + off, sizeof(line) - off,
+					"%s%s=%lu", off ? " " : "",
+					hook_names[i], hook_counts[i]);
+		if (off >= sizeof(line) - 64)
+			break;
+	}
+	if (any)
+		log_debug("hook call counts: %s", line);
+}
+
+/* ------------------------------------------------------------------ */
 /* Hooked entry points                                                 */
 /* ------------------------------------------------------------------ */
 
@@ -366,6 +447,7 @@ nvshipError_t hipMalloc(void **ptr, size_t size)
 	nvshipError_t r;
 
 	BOOTSTRAP();
+	BUMP(H_hipMalloc);
 	CHECK_REAL(hipMalloc);
 	if (ptr == NULL)
 		return NVSHIP_ERROR_INVALID_VALUE;
@@ -397,6 +479,7 @@ nvshipError_t hipExtMallocWithFlags(void **ptr, size_t size,
 				    unsigned int flags)
 {
 	BOOTSTRAP();
+	BUMP(H_hipExtMallocWithFlags);
 	if (disable_um && real.hipExtMallocWithFlags != NULL)
 		return real.hipExtMallocWithFlags(ptr, size, flags);
 	/* All flag variants become managed; the flags are advisory. */
@@ -410,6 +493,7 @@ nvshipError_t hipMallocAsync(void **ptr, size_t size, nvship_stream_t s)
 	 * pointer is valid earlier than required, which is safe. */
 	(void)s;
 	BOOTSTRAP();
+	BUMP(H_hipMallocAsync);
 	if (disable_um) {
 		CHECK_REAL(hipMallocAsync);
 		return real.hipMallocAsync(ptr, size, s);
@@ -428,6 +512,8 @@ nvshipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
 nvshipError_t hipFree(void *ptr)
 {
 	BOOTSTRAP();
+	BUMP(H_hipFree);
+	BUMP(H_hipMallocFromPoolAsync);
 	CHECK_REAL(hipFree);
 	if (ptr != NULL)
 		untrack_alloc(ptr);
@@ -437,6 +523,7 @@ nvshipError_t hipFree(void *ptr)
 nvshipError_t hipFreeAsync(void *ptr, nvship_stream_t stream)
 {
 	BOOTSTRAP();
+	BUMP(H_hipFreeAsync);
 	if (ptr == NULL)
 		return NVSHIP_SUCCESS;
 	if (untrack_alloc(ptr) != 0) {
@@ -456,6 +543,7 @@ nvshipError_t hipMemGetInfo(size_t *free_p, size_t *total_p)
 	size_t limit, freeb;
 
 	BOOTSTRAP();
+	BUMP(H_hipMemGetInfo);
 	CHECK_REAL(hipMemGetInfo);
 	if (disable_um)
 		return real.hipMemGetInfo(free_p, total_p);
@@ -473,6 +561,7 @@ nvshipError_t hipMemGetInfo(size_t *free_p, size_t *total_p)
 nvshipError_t hipSetDevice(int dev)
 {
 	BOOTSTRAP();
+	BUMP(H_hipSetDevice);
 	CHECK_REAL(hipSetDevice);
 	nvs_app_device = dev;
 	return real.hipSetDevice(dev);
@@ -506,6 +595,7 @@ nvshipError_t hipLaunchKernel(const void *f, nvship_dim3 grid,
 			      size_t shmem, nvship_stream_t stream)
 {
 	BOOTSTRAP();
+	BUMP(H_hipLaunchKernel);
 	CHECK_REAL(hipLaunchKernel);
 	GATED_LAUNCH(real.hipLaunchKernel(f, grid, block, args, shmem,
 					  stream));
@@ -518,6 +608,7 @@ nvshipError_t hipExtLaunchKernel(const void *f, nvship_dim3 grid,
 				 int flags)
 {
 	BOOTSTRAP();
+	BUMP(H_hipExtLaunchKernel);
 	CHECK_REAL(hipExtLaunchKernel);
 	GATED_LAUNCH(real.hipExtLaunchKernel(f, grid, block, args, shmem,
 					     stream, ev0, ev1, flags));
@@ -529,6 +620,7 @@ nvshipError_t hipLaunchCooperativeKernel(const void *f, nvship_dim3 grid,
 					 nvship_stream_t stream)
 {
 	BOOTSTRAP();
+	BUMP(H_hipLaunchCooperativeKernel);
 	CHECK_REAL(hipLaunchCooperativeKernel);
 	GATED_LAUNCH(real.hipLaunchCooperativeKernel(f, grid, block, args,
 						     shmem, stream));
@@ -542,6 +634,7 @@ nvshipError_t hipModuleLaunchKernel(nvship_function_t f, unsigned int gx,
 				    void **extra)
 {
 	BOOTSTRAP();
+	BUMP(H_hipModuleLaunchKernel);
 	CHECK_REAL(hipModuleLaunchKernel);
 	GATED_LAUNCH(real.hipModuleLaunchKernel(f, gx, gy, gz, bx, by, bz,
 						shmem, stream, params,
@@ -558,6 +651,7 @@ nvshipError_t hipExtModuleLaunchKernel(nvship_function_t f, uint32_t gwx,
 				       nvship_event_t ev1, uint32_t flags)
 {
 	BOOTSTRAP();
+	BUMP(H_hipExtModuleLaunchKernel);
 	CHECK_REAL(hipExtModuleLaunchKernel);
 	GATED_LAUNCH(real.hipExtModuleLaunchKernel(f, gwx, gwy, gwz, lwx,
 						   lwy, lwz, shmem, stream,
@@ -568,6 +662,7 @@ nvshipError_t hipExtModuleLaunchKernel(nvship_function_t f, uint32_t gwx,
 nvshipError_t hipGraphLaunch(nvship_graphexec_t g, nvship_stream_t stream)
 {
 	BOOTSTRAP();
+	BUMP(H_hipGraphLaunch);
 	CHECK_REAL(hipGraphLaunch);
 	GATED_LAUNCH(real.hipGraphLaunch(g, stream));
 }
@@ -576,6 +671,7 @@ nvshipError_t hipMemcpy(void *dst, const void *src, size_t n,
 			nvship_memcpy_kind kind)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpy);
 	CHECK_REAL(hipMemcpy);
 	GATED(real.hipMemcpy(dst, src, n, kind));
 }
@@ -584,6 +680,7 @@ nvshipError_t hipMemcpyAsync(void *dst, const void *src, size_t n,
 			     nvship_memcpy_kind kind, nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyAsync);
 	CHECK_REAL(hipMemcpyAsync);
 	GATED(real.hipMemcpyAsync(dst, src, n, kind, s));
 }
@@ -593,6 +690,7 @@ nvshipError_t hipMemcpyWithStream(void *dst, const void *src, size_t n,
 				  nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyWithStream);
 	CHECK_REAL(hipMemcpyWithStream);
 	GATED(real.hipMemcpyWithStream(dst, src, n, kind, s));
 }
@@ -601,6 +699,7 @@ nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
 			    size_t n)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyHtoD);
 	CHECK_REAL(hipMemcpyHtoD);
 	GATED(real.hipMemcpyHtoD(dst, src, n));
 }
@@ -608,6 +707,7 @@ nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
 nvshipError_t hipMemcpyDtoH(void *dst, nvship_deviceptr_t src, size_t n)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyDtoH);
 	CHECK_REAL(hipMemcpyDtoH);
 	GATED(real.hipMemcpyDtoH(dst, src, n));
 }
@@ -616,6 +716,7 @@ nvshipError_t hipMemcpyDtoD(nvship_deviceptr_t dst, nvship_deviceptr_t src,
 			    size_t n)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyDtoD);
 	CHECK_REAL(hipMemcpyDtoD);
 	GATED(real.hipMemcpyDtoD(dst, src, n));
 }
@@ -624,6 +725,7 @@ nvshipError_t hipMemcpyHtoDAsync(nvship_deviceptr_t dst, const void *src,
 				 size_t n, nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyHtoDAsync);
 	CHECK_REAL(hipMemcpyHtoDAsync);
 	GATED(real.hipMemcpyHtoDAsync(dst, src, n, s));
 }
@@ -632,6 +734,7 @@ nvshipError_t hipMemcpyDtoHAsync(void *dst, nvship_deviceptr_t src,
 				 size_t n, nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyDtoHAsync);
 	CHECK_REAL(hipMemcpyDtoHAsync);
 	GATED(real.hipMemcpyDtoHAsync(dst, src, n, s));
 }
@@ -641,6 +744,7 @@ nvshipError_t hipMemcpyDtoDAsync(nvship_deviceptr_t dst,
 				 nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemcpyDtoDAsync);
 	CHECK_REAL(hipMemcpyDtoDAsync);
 	GATED(real.hipMemcpyDtoDAsync(dst, src, n, s));
 }
@@ -648,6 +752,7 @@ nvshipError_t hipMemcpyDtoDAsync(nvship_deviceptr_t dst,
 nvshipError_t hipMemset(void *dst, int value, size_t n)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemset);
 	CHECK_REAL(hipMemset);
 	GATED(real.hipMemset(dst, value, n));
 }
@@ -656,8 +761,104 @@ nvshipError_t hipMemsetAsync(void *dst, int value, size_t n,
 			     nvship_stream_t s)
 {
 	BOOTSTRAP();
+	BUMP(H_hipMemsetAsync);
 	CHECK_REAL(hipMemsetAsync);
 	GATED(real.hipMemsetAsync(dst, value, n, s));
+}
+
+nvshipError_t hipMemsetD32Async(nvship_deviceptr_t dst, int value,
+				size_t count, nvship_stream_t s)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemsetD32Async);
+	CHECK_REAL(hipMemsetD32Async);
+	GATED(real.hipMemsetD32Async(dst, value, count, s));
+}
+
+nvshipError_t hipMemcpy2D(void *dst, size_t dpitch, const void *src,
+			  size_t spitch, size_t width, size_t height,
+			  nvship_memcpy_kind kind)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemcpy2D);
+	CHECK_REAL(hipMemcpy2D);
+	GATED(real.hipMemcpy2D(dst, dpitch, src, spitch, width, height,
+			       kind));
+}
+
+nvshipError_t hipMemcpy2DAsync(void *dst, size_t dpitch, const void *src,
+			       size_t spitch, size_t width, size_t height,
+			       nvship_memcpy_kind kind, nvship_stream_t s)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemcpy2DAsync);
+	CHECK_REAL(hipMemcpy2DAsync);
+	GATED(real.hipMemcpy2DAsync(dst, dpitch, src, spitch, width,
+				    height, kind, s));
+}
+
+nvshipError_t hipMemcpyToSymbol(const void *symbol, const void *src,
+				size_t n, size_t offset,
+				nvship_memcpy_kind kind)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemcpyToSymbol);
+	CHECK_REAL(hipMemcpyToSymbol);
+	GATED(real.hipMemcpyToSymbol(symbol, src, n, offset, kind));
+}
+
+nvshipError_t hipMemcpyFromSymbol(void *dst, const void *symbol, size_t n,
+				  size_t offset, nvship_memcpy_kind kind)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemcpyFromSymbol);
+	CHECK_REAL(hipMemcpyFromSymbol);
+	GATED(real.hipMemcpyFromSymbol(dst, symbol, n, offset, kind));
+}
+
+nvshipError_t hipMemcpyPeerAsync(void *dst, int dst_dev, const void *src,
+				 int src_dev, size_t n, nvship_stream_t s)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemcpyPeerAsync);
+	CHECK_REAL(hipMemcpyPeerAsync);
+	GATED(real.hipMemcpyPeerAsync(dst, dst_dev, src, src_dev, n, s));
+}
+
+nvshipError_t hipModuleLaunchCooperativeKernel(
+	nvship_function_t f, unsigned int gx, unsigned int gy,
+	unsigned int gz, unsigned int bx, unsigned int by, unsigned int bz,
+	unsigned int shmem, nvship_stream_t stream, void **params)
+{
+	BOOTSTRAP();
+	BUMP(H_hipModuleLaunchCooperativeKernel);
+	CHECK_REAL(hipModuleLaunchCooperativeKernel);
+	GATED_LAUNCH(real.hipModuleLaunchCooperativeKernel(
+		f, gx, gy, gz, bx, by, bz, shmem, stream, params));
+}
+
+/* Direct managed allocations by the app: tracked and capped too. */
+nvshipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags)
+{
+	nvshipError_t r;
+
+	BOOTSTRAP();
+	BUMP(H_hipMallocManaged);
+	CHECK_REAL(hipMallocManaged);
+	if (ptr == NULL)
+		return NVSHIP_ERROR_INVALID_VALUE;
+	if (!disable_um && !oversub_allowed && size > 0) {
+		pthread_mutex_lock(&alloc_mutex);
+		if (sum_allocated + size > mem_limit()) {
+			pthread_mutex_unlock(&alloc_mutex);
+			return NVSHIP_ERROR_OOM;
+		}
+		pthread_mutex_unlock(&alloc_mutex);
+	}
+	r = real.hipMallocManaged(ptr, size, flags);
+	if (r == NVSHIP_SUCCESS && size > 0 && !disable_um)
+		track_alloc(*ptr, size);
+	return r;
 }
 
 /* ------------------------------------------------------------------ */
@@ -696,6 +897,15 @@ static const struct hook_entry hook_table[] = {
 	{ "hipMemcpyDtoDAsync", (void *)hipMemcpyDtoDAsync },
 	{ "hipMemset", (void *)hipMemset },
 	{ "hipMemsetAsync", (void *)hipMemsetAsync },
+	{ "hipMemsetD32Async", (void *)hipMemsetD32Async },
+	{ "hipMemcpy2D", (void *)hipMemcpy2D },
+	{ "hipMemcpy2DAsync", (void *)hipMemcpy2DAsync },
+	{ "hipMemcpyToSymbol", (void *)hipMemcpyToSymbol },
+	{ "hipMemcpyFromSymbol", (void *)hipMemcpyFromSymbol },
+	{ "hipMemcpyPeerAsync", (void *)hipMemcpyPeerAsync },
+	{ "hipModuleLaunchCooperativeKernel",
+	  (void *)hipModuleLaunchCooperativeKernel },
+	{ "hipMallocManaged", (void *)hipMallocManaged },
 	{ NULL, NULL },
 };
 
@@ -719,6 +929,7 @@ nvshipError_t hipGetProcAddress(const char *symbol, void **pfn,
 	void *ours;
 
 	BOOTSTRAP();
+	BUMP(H_hipGetProcAddress);
 	CHECK_REAL(hipGetProcAddress);
 	r = real.hipGetProcAddress(symbol, pfn, hip_version, flags,
 				   symbol_status);
