@@ -142,9 +142,14 @@ def main(argv=None):
     amp_dtype = getattr(torch, dtype) if dtype != "float32" else None
     torch.manual_seed(1234 + rank)
     model = build(model_name, 1000 if device == "cuda" else 10).to(dev)
+    if device == "cuda":
+        torch.backends.cudnn.benchmark = True  # MIOpen autotune
+        model = model.to(memory_format=torch.channels_last)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
     lossf = torch.nn.CrossEntropyLoss()
     x = torch.randn(batch, 3, image, image, device=dev)
+    if device == "cuda":
+        x = x.to(memory_format=torch.channels_last)
     y = torch.randint(0, model.fc.out_features, (batch,), device=dev)
 
     def step():

@@ -33,10 +33,18 @@ def run_training(model_name: str = "resnet50", device: str = "cuda",
     dev = torch.device(device)
     amp_dtype = getattr(torch, dtype) if dtype != "float32" else None
     torch.manual_seed(0)
-    model = build(model_name, num_classes).to(dev)
+    if device.startswith("cuda"):
+        # MIOpen autotune + NHWC: the fast conv path on CDNA.
+        torch.backends.cudnn.benchmark = True
+        model = build(model_name, num_classes).to(
+            dev, memory_format=torch.channels_last)
+    else:
+        model = build(model_name, num_classes).to(dev)
     opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
     lossf = torch.nn.CrossEntropyLoss()
     x = torch.randn(batch, 3, image, image, device=dev)
+    if device.startswith("cuda"):
+        x = x.to(memory_format=torch.channels_last)
     y = torch.randint(0, num_classes, (batch,), device=dev)
 
     def step():

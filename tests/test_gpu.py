@@ -89,7 +89,8 @@ def test_interposition_caps_torch_alloc(sched, sock_dir):
     """With a small fake total and no oversub, a big torch alloc OOMs —
     proof the interposer intercepts torch's allocator."""
     code = (
-        "import torch; torch.cuda.init(); "
+        "import torch\n"
+        "torch.cuda.init()\n"
         "try:\n"
         "    x = torch.empty(1024, 1024, 1024, device='cuda')  # 4 GiB\n"
         "    print('ALLOC_SUCCEEDED')\n"
