@@ -107,6 +107,7 @@ class Client:
     sock: socket.socket | None = None
     client_id: int = 0
     scheduling_on: bool = field(default=True)
+    gpu: int = 0
 
     def connect(self, timeout: float = 10.0) -> "Client":
         self.sock = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
@@ -117,7 +118,8 @@ class Client:
     def register(self, timeout: float = 10.0) -> Message:
         assert self.sock is not None
         send_msg(self.sock, Message(REGISTER, self.pod_name,
-                                    self.pod_namespace))
+                                    self.pod_namespace,
+                                    data=f"gpu{self.gpu}"))
         reply = recv_msg(self.sock, timeout)
         if reply.type not in (SCHED_ON, SCHED_OFF):
             raise RuntimeError(f"unexpected handshake: {reply.type_name}")

@@ -66,6 +66,23 @@ static int sock_fd = -1;
 static sem_t init_done_sem;
 static char pod_name[NVS_POD_NAME_LEN];
 static char pod_namespace[NVS_POD_NS_LEN];
+static int physical_gpu;  /* node-level GPU index this client shares */
+
+/* The node-level index of the GPU this process uses: NVSHARE_GPU, or
+ * the first entry of ROCR/HIP_VISIBLE_DEVICES (containers see exactly
+ * one GPU), else 0. */
+static int detect_physical_gpu(void)
+{
+	const char *v = getenv("NVSHARE_GPU");
+
+	if (v == NULL)
+		v = getenv("ROCR_VISIBLE_DEVICES");
+	if (v == NULL)
+		v = getenv("HIP_VISIBLE_DEVICES");
+	if (v != NULL && v[0] >= '0' && v[0] <= '9')
+		return atoi(v);
+	return 0;
+}
 
 /* idle detection */
 static fn_rsmi_init p_rsmi_init;
@@ -180,7 +197,12 @@ static int connect_and_register(void)
 	fd = nvs_connect_path(path);
 	if (fd < 0)
 		return -1;
-	nvs_msg_init(&m, NVS_REGISTER, 0, NULL);
+	{
+		char gpu_tag[NVS_MSG_DATA_LEN];
+
+		snprintf(gpu_tag, sizeof(gpu_tag), "gpu%d", physical_gpu);
+		nvs_msg_init(&m, NVS_REGISTER, 0, gpu_tag);
+	}
 	nvs_strlcpy(m.pod_name, pod_name, sizeof(m.pod_name));
 	nvs_strlcpy(m.pod_namespace, pod_namespace, sizeof(m.pod_namespace));
 	if (nvs_send_msg(fd, &m) != 0) {
@@ -318,14 +340,8 @@ static void *client_thread(void *arg)
 static void idle_detect_init(void)
 {
 	void *h;
-	const char *v;
 
-	/* Which physical device are we? First entry of the visible list. */
-	v = getenv("ROCR_VISIBLE_DEVICES");
-	if (v == NULL)
-		v = getenv("HIP_VISIBLE_DEVICES");
-	if (v != NULL && v[0] >= '0' && v[0] <= '9')
-		rsmi_dev_index = (uint32_t)atoi(v);
+	rsmi_dev_index = (uint32_t)physical_gpu;
 
 	h = dlopen("librocm_smi64.so.7", RTLD_LAZY | RTLD_LOCAL);
 	if (h == NULL)
@@ -422,6 +438,7 @@ void nvs_client_init(void)
 
 	sem_init(&init_done_sem, 0, 0);
 	read_pod_identity();
+	physical_gpu = detect_physical_gpu();
 
 	if (standalone) {
 		scheduler_on = 0;

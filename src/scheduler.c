@@ -2,20 +2,26 @@
  * nvshare-scheduler — per-node GPU arbiter for nvshare-amd (MI355X).
  *
  * Serializes GPU work across transparent-sharing clients with an FCFS
- * exclusive lock held for a time quantum (TQ).  Single-threaded epoll
- * event loop plus one timer thread.  Wire-compatible with the reference
- * protocol (see proto.h); behavioral parity reference:
- * /root/reference/src/scheduler.c (registry, FCFS queue, TQ timer,
- * strict eviction, SCHED_ON/OFF broadcast, SET_TQ).
+ * exclusive lock held for a time quantum (TQ), independently for every
+ * GPU on the node.  Single-threaded epoll event loop plus one timer
+ * thread.  Wire-compatible with the reference protocol (see proto.h);
+ * behavioral parity reference: /root/reference/src/scheduler.c
+ * (registry, FCFS queue, TQ timer, strict eviction, SCHED_ON/OFF
+ * broadcast, SET_TQ).
  *
  * nvshare-amd additions over the reference:
+ *   - multi-GPU: one daemon arbitrates all 8 GPUs of an MI355X node
+ *     with independent per-GPU locks/queues; clients declare their GPU
+ *     in REGISTER.data ("gpuN"; empty = GPU 0, so reference-shaped
+ *     clients still work).  The reference hard-coded a single GPU
+ *     (README.md:97);
  *   - env-configurable startup state: NVSHARE_TQ (seconds),
  *     NVSHARE_SCHED_OFF=1 (start with scheduling disabled)
  *     (reference left these as TODOs, scheduler.c:549-552);
  *   - solo-client fast path: the TQ timer does not preempt the lock
- *     holder when nobody else is queued, so a lone client never pays
- *     periodic drain stalls; a newly arriving waiter preempts the
- *     holder as soon as the holder has had >= one full quantum;
+ *     holder when nobody else is queued on that GPU, so a lone client
+ *     never pays periodic drain stalls; a newly arriving waiter
+ *     preempts the holder as soon as it has had >= one full quantum;
  *   - STATUS_REQ/STATUS query for observability (nvsharectl -q);
  *   - partial-read tolerant framing (per-connection receive buffer);
  *   - SIGUSR1 dumps scheduler state to stderr.
@@ -39,49 +45,65 @@
 
 #define NVS_DEFAULT_TQ 30
 #define MAX_EPOLL_EVENTS 64
+#define MAX_GPUS 16
 
-/* One TCP^WUnix connection. Becomes a "client" after REGISTER. */
+/* One Unix-socket connection. Becomes a "client" after REGISTER. */
 struct conn {
 	int fd;
 	size_t got;               /* bytes of in-progress message */
 	struct nvs_msg inmsg;
 	int registered;
 	uint64_t id;              /* client id (registered only) */
+	int gpu;                  /* GPU this client shares */
 	char pod_name[NVS_POD_NAME_LEN];
 	char pod_namespace[NVS_POD_NS_LEN];
-	int wants_lock;           /* present in request queue */
+	int wants_lock;           /* present in its GPU's request queue */
 	struct conn *next;        /* registry list */
 	struct conn *qnext;       /* FCFS queue list */
 };
 
-static struct conn *clients;      /* all connections (incl. unregistered) */
-static struct conn *queue_head;   /* FCFS lock queue */
-static struct conn *queue_tail;
+/* Per-GPU arbitration state. */
+struct gpu_state {
+	struct conn *queue_head;  /* FCFS queue; head == holder */
+	struct conn *queue_tail;
+	int lock_held;
+	struct conn *lock_holder;
+	int drop_lock_sent;
+	unsigned long round;
+	int64_t quantum_start_ns;
+	unsigned long grants, preemptions;
+};
+
+static struct conn *clients;      /* all connections */
+static struct gpu_state gpus[MAX_GPUS];
 
 static pthread_mutex_t g_mutex = PTHREAD_MUTEX_INITIALIZER;
 static pthread_cond_t timer_cv = PTHREAD_COND_INITIALIZER;
 
 static int scheduler_on = 1;
 static int tq_seconds = NVS_DEFAULT_TQ;
-static int lock_held;
-static struct conn *lock_holder;
-static int drop_lock_sent;
-static unsigned long scheduling_round;
-static int64_t quantum_start_ns;
-/* stats */
-static unsigned long total_grants, total_preemptions, total_evictions;
+static unsigned long total_evictions;
 
 static int epoll_fd = -1;
 static char sock_path[NVS_SOCK_PATH_MAX];
 static volatile sig_atomic_t dump_requested;
 
-static int queue_len(void)
+static int queue_len(const struct gpu_state *g)
 {
 	int n = 0;
 	struct conn *c;
 
-	for (c = queue_head; c != NULL; c = c->qnext)
+	for (c = g->queue_head; c != NULL; c = c->qnext)
 		n++;
+	return n;
+}
+
+static int queue_len_all(void)
+{
+	int n = 0, i;
+
+	for (i = 0; i < MAX_GPUS; i++)
+		n += queue_len(&gpus[i]);
 	return n;
 }
 
@@ -109,27 +131,27 @@ static void send_to(struct conn *c, uint8_t type, const char *data)
 			  nvs_msg_type_str(type), c->id);
 }
 
-static void queue_push(struct conn *c)
+static void queue_push(struct gpu_state *g, struct conn *c)
 {
 	c->qnext = NULL;
-	if (queue_tail != NULL)
-		queue_tail->qnext = c;
+	if (g->queue_tail != NULL)
+		g->queue_tail->qnext = c;
 	else
-		queue_head = c;
-	queue_tail = c;
+		g->queue_head = c;
+	g->queue_tail = c;
 	c->wants_lock = 1;
 }
 
-static void queue_remove(struct conn *c)
+static void queue_remove(struct gpu_state *g, struct conn *c)
 {
-	struct conn **pp = &queue_head;
+	struct conn **pp = &g->queue_head;
 	struct conn *prev = NULL, *it;
 
-	for (it = queue_head; it != NULL; prev = it, it = it->qnext) {
+	for (it = g->queue_head; it != NULL; prev = it, it = it->qnext) {
 		if (it == c) {
 			*pp = it->qnext;
-			if (queue_tail == c)
-				queue_tail = prev;
+			if (g->queue_tail == c)
+				g->queue_tail = prev;
 			break;
 		}
 		pp = &it->qnext;
@@ -138,79 +160,90 @@ static void queue_remove(struct conn *c)
 	c->wants_lock = 0;
 }
 
-/* Grant the lock to the queue head if possible. Caller holds g_mutex. */
-static void try_schedule(void)
+/* Grant GPU g's lock to its queue head if possible. Holds g_mutex. */
+static void try_schedule(struct gpu_state *g)
 {
-	if (!scheduler_on || lock_held || queue_head == NULL)
+	if (!scheduler_on || g->lock_held || g->queue_head == NULL)
 		return;
-	lock_holder = queue_head;
-	lock_held = 1;
-	drop_lock_sent = 0;
-	scheduling_round++;
-	total_grants++;
-	quantum_start_ns = nvs_now_ns();
-	send_to(lock_holder, NVS_LOCK_OK, NULL);
-	log_debug("round %lu: lock -> %016" PRIx64 " (queue=%d)",
-		  scheduling_round, lock_holder->id, queue_len());
+	g->lock_holder = g->queue_head;
+	g->lock_held = 1;
+	g->drop_lock_sent = 0;
+	g->round++;
+	g->grants++;
+	g->quantum_start_ns = nvs_now_ns();
+	send_to(g->lock_holder, NVS_LOCK_OK, NULL);
+	log_debug("gpu%ld round %lu: lock -> %016" PRIx64 " (queue=%d)",
+		  (long)(g - gpus), g->round, g->lock_holder->id,
+		  queue_len(g));
 	pthread_cond_broadcast(&timer_cv);
 }
 
+/* A GPU is "armed" when its holder can be preempted: lock held, at
+ * least one waiter behind the holder, DROP_LOCK not yet sent. */
+static int gpu_armed(const struct gpu_state *g)
+{
+	return g->lock_held && scheduler_on && !g->drop_lock_sent &&
+	       g->queue_head != NULL && g->queue_head->qnext != NULL;
+}
+
 /*
- * TQ timer thread.  While a client holds the lock and at least one other
- * client is queued, send DROP_LOCK once the holder has had tq seconds.
- * A lone holder is never preempted (solo fast path); the arrival of a
- * waiter wakes this thread, which then preempts as soon as the holder's
- * cumulative quantum is exhausted.  After DROP_LOCK we wait indefinitely
- * for LOCK_RELEASED; only socket death evicts a stuck client (strict
- * eviction, as the reference: scheduler.c:352,645-663).
+ * TQ timer thread.  For every armed GPU, send DROP_LOCK once the
+ * holder has had tq seconds.  A lone holder is never preempted (solo
+ * fast path); a waiter's arrival wakes this thread, which preempts as
+ * soon as the holder's quantum is exhausted.  After DROP_LOCK we wait
+ * indefinitely for LOCK_RELEASED; only socket death evicts a stuck
+ * client (strict eviction, as the reference: scheduler.c:352,645-663).
  */
 static void *timer_thread(void *arg)
 {
 	(void)arg;
 	pthread_mutex_lock(&g_mutex);
 	for (;;) {
-		while (!lock_held || !scheduler_on || drop_lock_sent ||
-		       queue_head == NULL || queue_head->qnext == NULL) {
-			pthread_cond_wait(&timer_cv, &g_mutex);
-		}
-		/* Holder + >=1 waiter: arm the deadline. */
-		{
-			unsigned long round = scheduling_round;
-			int64_t deadline_ns =
-				quantum_start_ns +
-				(int64_t)tq_seconds * 1000000000LL;
-			int64_t now = nvs_now_ns();
+		int64_t now = nvs_now_ns();
+		int64_t next_deadline = INT64_MAX;
+		int i, any_armed = 0;
 
-			if (now < deadline_ns) {
-				struct timespec abs;
-				clock_gettime(CLOCK_REALTIME, &abs);
-				int64_t wait_ns = deadline_ns - now;
-				abs.tv_sec += wait_ns / 1000000000LL;
-				abs.tv_nsec += wait_ns % 1000000000LL;
-				if (abs.tv_nsec >= 1000000000L) {
-					abs.tv_sec++;
-					abs.tv_nsec -= 1000000000L;
-				}
-				pthread_cond_timedwait(&timer_cv, &g_mutex,
-						       &abs);
-				/* State may have changed; re-evaluate. */
-				if (scheduling_round != round)
-					continue;
-				if (nvs_now_ns() <
-				    quantum_start_ns +
-				    (int64_t)tq_seconds * 1000000000LL)
-					continue;
-			}
-			if (scheduling_round != round || !lock_held ||
-			    !scheduler_on || drop_lock_sent ||
-			    queue_head == NULL || queue_head->qnext == NULL)
+		for (i = 0; i < MAX_GPUS; i++) {
+			struct gpu_state *g = &gpus[i];
+			int64_t deadline;
+
+			if (!gpu_armed(g))
 				continue;
-			drop_lock_sent = 1;
-			total_preemptions++;
-			log_debug("round %lu: TQ expired, DROP_LOCK -> "
-				  "%016" PRIx64, scheduling_round,
-				  lock_holder->id);
-			send_to(lock_holder, NVS_DROP_LOCK, NULL);
+			any_armed = 1;
+			deadline = g->quantum_start_ns +
+				   (int64_t)tq_seconds * 1000000000LL;
+			if (deadline <= now) {
+				g->drop_lock_sent = 1;
+				g->preemptions++;
+				log_debug("gpu%d round %lu: TQ expired, "
+					  "DROP_LOCK -> %016" PRIx64, i,
+					  g->round, g->lock_holder->id);
+				send_to(g->lock_holder, NVS_DROP_LOCK,
+					NULL);
+				continue;
+			}
+			if (deadline < next_deadline)
+				next_deadline = deadline;
+		}
+
+		if (!any_armed || next_deadline == INT64_MAX) {
+			pthread_cond_wait(&timer_cv, &g_mutex);
+			continue;
+		}
+		{
+			struct timespec abs;
+			int64_t wait_ns = next_deadline - nvs_now_ns();
+
+			if (wait_ns <= 0)
+				continue;
+			clock_gettime(CLOCK_REALTIME, &abs);
+			abs.tv_sec += wait_ns / 1000000000LL;
+			abs.tv_nsec += wait_ns % 1000000000LL;
+			if (abs.tv_nsec >= 1000000000L) {
+				abs.tv_sec++;
+				abs.tv_nsec -= 1000000000L;
+			}
+			pthread_cond_timedwait(&timer_cv, &g_mutex, &abs);
 		}
 	}
 	return NULL;
@@ -229,6 +262,7 @@ static void bcast_status(void)
 static void delete_conn(struct conn *c)
 {
 	struct conn **pp;
+	struct gpu_state *g = &gpus[c->gpu];
 
 	epoll_ctl(epoll_fd, EPOLL_CTL_DEL, c->fd, NULL);
 	close(c->fd);
@@ -240,11 +274,11 @@ static void delete_conn(struct conn *c)
 			 c->pod_name[0] ? c->pod_name : "-");
 	}
 	if (c->wants_lock)
-		queue_remove(c);
-	if (lock_holder == c) {
-		lock_held = 0;
-		lock_holder = NULL;
-		drop_lock_sent = 0;
+		queue_remove(g, c);
+	if (g->lock_holder == c) {
+		g->lock_held = 0;
+		g->lock_holder = NULL;
+		g->drop_lock_sent = 0;
 	}
 	for (pp = &clients; *pp != NULL; pp = &(*pp)->next) {
 		if (*pp == c) {
@@ -253,7 +287,7 @@ static void delete_conn(struct conn *c)
 		}
 	}
 	free(c);
-	try_schedule();
+	try_schedule(g);
 }
 
 static void handle_status_req(struct conn *c)
@@ -261,17 +295,33 @@ static void handle_status_req(struct conn *c)
 	struct nvs_msg m;
 	char buf[NVS_MSG_DATA_LEN];
 
-	/* data: "<on>,<tq>,<nclients>,<qlen>" — fits in 20 bytes for sane
-	 * values; counts are clamped by the field width. */
+	/* data: "<on>,<tq>,<nclients>,<qlen>" */
 	snprintf(buf, sizeof(buf), "%d,%d,%d,%d", scheduler_on, tq_seconds,
-		 client_count(), queue_len());
+		 client_count(), queue_len_all());
 	nvs_msg_init(&m, NVS_STATUS, c->id, buf);
 	if (nvs_send_msg(c->fd, &m) != 0)
 		log_warn("STATUS reply failed");
 }
 
+static int parse_gpu_index(const struct nvs_msg *m)
+{
+	char buf[NVS_MSG_DATA_LEN];
+	long v;
+
+	memcpy(buf, m->data, NVS_MSG_DATA_LEN);
+	buf[NVS_MSG_DATA_LEN - 1] = '\0';
+	if (strncmp(buf, "gpu", 3) != 0)
+		return 0;
+	v = strtol(buf + 3, NULL, 10);
+	if (v < 0 || v >= MAX_GPUS)
+		return 0;
+	return (int)v;
+}
+
 static void process_msg(struct conn *c, const struct nvs_msg *m)
 {
+	struct gpu_state *g = &gpus[c->gpu];
+
 	log_debug("recv %s from fd=%d id=%016" PRIx64,
 		  nvs_msg_type_str(m->type), c->fd, c->id);
 
@@ -285,14 +335,16 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 		}
 		c->registered = 1;
 		c->id = nvs_gen_id();
+		c->gpu = parse_gpu_index(m);
 		memcpy(c->pod_name, m->pod_name, NVS_POD_NAME_LEN);
 		c->pod_name[NVS_POD_NAME_LEN - 1] = '\0';
 		memcpy(c->pod_namespace, m->pod_namespace, NVS_POD_NS_LEN);
 		c->pod_namespace[NVS_POD_NS_LEN - 1] = '\0';
 		snprintf(idbuf, sizeof(idbuf), "%016" PRIx64, c->id);
-		log_info("registered client %016" PRIx64 " (%s/%s)", c->id,
+		log_info("registered client %016" PRIx64 " (%s/%s) on gpu%d",
+			 c->id,
 			 c->pod_namespace[0] ? c->pod_namespace : "-",
-			 c->pod_name[0] ? c->pod_name : "-");
+			 c->pod_name[0] ? c->pod_name : "-", c->gpu);
 		send_to(c, scheduler_on ? NVS_SCHED_ON : NVS_SCHED_OFF,
 			idbuf);
 		break;
@@ -302,31 +354,29 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 			log_warn("REQ_LOCK from unregistered fd=%d", c->fd);
 			break;
 		}
-		if (!scheduler_on) {
-			/* Clients free-run while scheduling is off. */
+		if (!scheduler_on)
+			break; /* clients free-run while scheduling is off */
+		if (c->wants_lock || g->lock_holder == c)
 			break;
-		}
-		if (c->wants_lock || lock_holder == c)
-			break;
-		queue_push(c);
-		if (!lock_held)
-			try_schedule();
+		queue_push(g, c);
+		if (!g->lock_held)
+			try_schedule(g);
 		else
-			pthread_cond_broadcast(&timer_cv); /* arm preemption */
+			pthread_cond_broadcast(&timer_cv); /* arm preempt */
 		break;
 	case NVS_LOCK_RELEASED:
-		if (lock_holder != c) {
-			/* Late/duplicate release (e.g. early-release racing
-			 * a preemption) — ignore. */
+		if (g->lock_holder != c) {
+			/* Late/duplicate release (early-release racing a
+			 * preemption) — ignore. */
 			log_debug("stale LOCK_RELEASED from %016" PRIx64,
 				  c->id);
 			break;
 		}
-		queue_remove(c);
-		lock_held = 0;
-		lock_holder = NULL;
-		drop_lock_sent = 0;
-		try_schedule();
+		queue_remove(g, c);
+		g->lock_held = 0;
+		g->lock_holder = NULL;
+		g->drop_lock_sent = 0;
+		try_schedule(g);
 		break;
 	case NVS_SET_TQ: {
 		char buf[NVS_MSG_DATA_LEN];
@@ -347,29 +397,37 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 	}
 	case NVS_SCHED_ON:
 		if (!scheduler_on) {
+			int i;
+
 			scheduler_on = 1;
 			log_info("scheduling enabled");
 			bcast_status();
-			try_schedule();
+			for (i = 0; i < MAX_GPUS; i++)
+				try_schedule(&gpus[i]);
 		}
 		break;
 	case NVS_SCHED_OFF:
 		if (scheduler_on) {
-			struct conn *it, *nx;
+			int i;
 
 			scheduler_on = 0;
 			log_info("scheduling disabled (free-for-all)");
 			bcast_status();
-			/* Flush the queue; everyone may run. */
-			for (it = queue_head; it != NULL; it = nx) {
-				nx = it->qnext;
-				it->qnext = NULL;
-				it->wants_lock = 0;
+			for (i = 0; i < MAX_GPUS; i++) {
+				struct gpu_state *gi = &gpus[i];
+				struct conn *it, *nx;
+
+				for (it = gi->queue_head; it != NULL;
+				     it = nx) {
+					nx = it->qnext;
+					it->qnext = NULL;
+					it->wants_lock = 0;
+				}
+				gi->queue_head = gi->queue_tail = NULL;
+				gi->lock_held = 0;
+				gi->lock_holder = NULL;
+				gi->drop_lock_sent = 0;
 			}
-			queue_head = queue_tail = NULL;
-			lock_held = 0;
-			lock_holder = NULL;
-			drop_lock_sent = 0;
 		}
 		break;
 	case NVS_STATUS_REQ:
@@ -410,17 +468,28 @@ static int handle_readable(struct conn *c)
 static void dump_state(void)
 {
 	struct conn *c;
+	int i;
 
 	log_info("=== scheduler state ===");
-	log_info("on=%d tq=%ds lock_held=%d holder=%016" PRIx64
-		 " round=%lu grants=%lu preempts=%lu evictions=%lu",
-		 scheduler_on, tq_seconds, lock_held,
-		 lock_holder ? lock_holder->id : 0, scheduling_round,
-		 total_grants, total_preemptions, total_evictions);
+	log_info("on=%d tq=%ds clients=%d evictions=%lu", scheduler_on,
+		 tq_seconds, client_count(), total_evictions);
+	for (i = 0; i < MAX_GPUS; i++) {
+		struct gpu_state *g = &gpus[i];
+
+		if (!g->grants && !queue_len(g))
+			continue;
+		log_info(" gpu%d: lock_held=%d holder=%016" PRIx64
+			 " round=%lu grants=%lu preempts=%lu queue=%d", i,
+			 g->lock_held,
+			 g->lock_holder ? g->lock_holder->id : 0,
+			 g->round, g->grants, g->preemptions,
+			 queue_len(g));
+	}
 	for (c = clients; c != NULL; c = c->next)
-		log_info("  conn fd=%d reg=%d id=%016" PRIx64 " queued=%d%s",
-			 c->fd, c->registered, c->id, c->wants_lock,
-			 lock_holder == c ? " [HOLDER]" : "");
+		log_info("  conn fd=%d reg=%d id=%016" PRIx64 " gpu=%d "
+			 "queued=%d%s", c->fd, c->registered, c->id,
+			 c->gpu, c->wants_lock,
+			 gpus[c->gpu].lock_holder == c ? " [HOLDER]" : "");
 }
 
 static void on_sigusr1(int sig)
@@ -471,8 +540,8 @@ int main(void)
 		     == 0);
 
 	log_info("nvshare-scheduler (amd) listening on %s (tq=%ds, "
-		 "scheduling %s)", sock_path, tq_seconds,
-		 scheduler_on ? "on" : "off");
+		 "scheduling %s, up to %d GPUs)", sock_path, tq_seconds,
+		 scheduler_on ? "on" : "off", MAX_GPUS);
 
 	for (;;) {
 		int nev, i;
@@ -508,7 +577,6 @@ int main(void)
 
 					if (fd < 0)
 						break;
-					/* Nonblocking for partial reads. */
 					{
 						int fl = fcntl(fd, F_GETFL);
 						fcntl(fd, F_SETFL,

@@ -14,8 +14,8 @@ import pytest
 from nvshare_amd import ctl, proto
 
 
-def make_client(sock_dir, name="c"):
-    c = proto.Client(sock_dir=sock_dir, pod_name=name)
+def make_client(sock_dir, name="c", gpu=0):
+    c = proto.Client(sock_dir=sock_dir, pod_name=name, gpu=gpu)
     c.connect()
     c.register()
     return c
@@ -177,6 +177,43 @@ def test_stale_lock_released_ignored(sched, sock_dir):
     assert a.recv(5).type == proto.LOCK_OK
     a.close()
     b.close()
+
+
+def test_multi_gpu_independent_locks(sched, sock_dir):
+    """Clients on different GPUs hold their locks CONCURRENTLY; the
+    reference was single-GPU (README.md:97) — nvshare-amd arbitrates
+    each of a node's 8 MI355Xs independently."""
+    a = make_client(sock_dir, "a", gpu=0)
+    b = make_client(sock_dir, "b", gpu=1)
+    a.send(proto.REQ_LOCK)
+    b.send(proto.REQ_LOCK)
+    assert a.recv(5).type == proto.LOCK_OK
+    assert b.recv(5).type == proto.LOCK_OK  # no waiting on gpu0's lock
+    # Neither gets preempted (each is solo on its GPU, TQ=1).
+    with pytest.raises(Exception):
+        a.recv(2.0)
+    a.close()
+    b.close()
+
+
+def test_multi_gpu_preemption_is_per_gpu(sched, sock_dir):
+    """A waiter on gpu1 must not preempt the holder on gpu0."""
+    a = make_client(sock_dir, "a", gpu=0)
+    b = make_client(sock_dir, "b", gpu=1)
+    c = make_client(sock_dir, "c", gpu=1)
+    a.send(proto.REQ_LOCK)
+    assert a.recv(5).type == proto.LOCK_OK
+    b.send(proto.REQ_LOCK)
+    assert b.recv(5).type == proto.LOCK_OK
+    c.send(proto.REQ_LOCK)  # waits behind b on gpu1
+    # b gets DROP_LOCK after ~TQ; a (solo on gpu0) must not.
+    assert b.recv(5).type == proto.DROP_LOCK
+    b.send(proto.LOCK_RELEASED)
+    assert c.recv(5).type == proto.LOCK_OK
+    with pytest.raises(Exception):
+        a.recv(1.0)
+    for x in (a, b, c):
+        x.close()
 
 
 def test_env_tq_startup(artifacts, sock_dir):
