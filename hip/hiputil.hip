@@ -146,6 +146,13 @@ int nvs_hip_prefetch(const void *p, size_t bytes, int device)
 	return 0;
 }
 
+/* advice: 3=SetPreferredLocation, 5=SetAccessedBy, 100=SetCoarseGrain */
+int nvs_hip_advise(const void *p, size_t bytes, int advice, int device)
+{
+	CHECK(hipMemAdvise(p, bytes, (hipMemoryAdvise)advice, device));
+	return 0;
+}
+
 int nvs_hip_sync(void)
 {
 	CHECK(hipDeviceSynchronize());

@@ -44,6 +44,16 @@ class HipUtil:
                                            ctypes.c_size_t(nbytes),
                                            device), "hipMemPrefetchAsync")
 
+    ADVISE_PREFERRED_LOCATION = 3
+    ADVISE_ACCESSED_BY = 5
+    ADVISE_COARSE_GRAIN = 100
+
+    def advise(self, ptr: int, nbytes: int, advice: int,
+               device: int = 0) -> None:
+        self._ck(self.lib.nvs_hip_advise(ctypes.c_void_p(ptr),
+                                         ctypes.c_size_t(nbytes),
+                                         advice, device), "hipMemAdvise")
+
     def touch_pages(self, ptr: int, n_floats: int, stride: int = 1,
                     val: float = 1.0) -> None:
         self._ck(self.lib.nvs_touch_pages(ctypes.c_void_p(ptr),
@@ -81,8 +91,9 @@ def load() -> HipUtil:
             f"{path} missing; build with `make -C hip` (hipcc, gfx950)")
     lib = ctypes.CDLL(str(path))
     for fname in ("nvs_hip_malloc_managed", "nvs_hip_malloc",
-                  "nvs_hip_free", "nvs_hip_prefetch", "nvs_touch_pages",
-                  "nvs_read_pages", "nvs_busy", "nvs_stream_triad",
-                  "nvs_hip_sync", "nvs_hip_mem_get_info"):
+                  "nvs_hip_free", "nvs_hip_prefetch", "nvs_hip_advise",
+                  "nvs_touch_pages", "nvs_read_pages", "nvs_busy",
+                  "nvs_stream_triad", "nvs_hip_sync",
+                  "nvs_hip_mem_get_info"):
         getattr(lib, fname).restype = ctypes.c_int
     return HipUtil(lib)
