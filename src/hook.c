@@ -92,6 +92,8 @@ static size_t mem_total;        /* advertised total (bytes) */
 static size_t mem_reserve;      /* carve-out (bytes) */
 static int oversub_allowed;
 static int disable_um;
+static int alloc_prefetch = 1;   /* NVSHARE_ALLOC_PREFETCH */
+static int coarse_grain = 1;     /* NVSHARE_COARSE_GRAIN */
 static pthread_once_t memquery_once = PTHREAD_ONCE_INIT;
 
 /* pending-kernel window */
@@ -269,8 +271,6 @@ static size_t mem_limit(void)
 /* Allocation tracking                                                 */
 /* ------------------------------------------------------------------ */
 
-static int alloc_prefetch = 1;   /* NVSHARE_ALLOC_PREFETCH */
-static int coarse_grain = 1;     /* NVSHARE_COARSE_GRAIN */
 #define NVSHIP_MEM_ADVISE_SET_COARSE_GRAIN 100
 
 /*
