@@ -1,0 +1,76 @@
+"""ThreadSanitizer run of the scheduler daemon (race detection).
+
+The reference had no sanitizer coverage; here the
+daemon's full lock lifecycle runs under TSan and the test fails on any
+reported race.
+"""
+
+from __future__ import annotations
+
+import os
+import signal
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+from nvshare_amd import proto
+
+REPO = Path(__file__).resolve().parent.parent
+TSAN_BIN = REPO / "src" / "build-tsan" / "nvshare-scheduler"
+
+
+@pytest.fixture(scope="module")
+def tsan_bin():
+    r = subprocess.run(["make", "-s", "tsan"], cwd=REPO,
+                       capture_output=True, text=True)
+    if r.returncode != 0 or not TSAN_BIN.exists():
+        pytest.skip(f"tsan build unavailable: {r.stderr[-500:]}")
+    return TSAN_BIN
+
+
+def test_scheduler_lifecycle_under_tsan(tsan_bin, sock_dir):
+    env = dict(os.environ)
+    env["NVSHARE_SOCK_DIR"] = sock_dir
+    env["NVSHARE_TQ"] = "1"
+    env["NVSHARE_DEBUG"] = "1"
+    env["TSAN_OPTIONS"] = "exitcode=66 halt_on_error=0"
+    log = open(os.path.join(sock_dir, "tsan.log"), "wb")
+    p = subprocess.Popen([str(tsan_bin)], env=env, stdout=log,
+                         stderr=log)
+    try:
+        deadline = time.monotonic() + 10
+        spath = proto.scheduler_path(sock_dir)
+        while not os.path.exists(spath):
+            assert time.monotonic() < deadline, "socket never appeared"
+            assert p.poll() is None
+            time.sleep(0.05)
+
+        # Drive a full multi-client lifecycle incl. TQ preemption.
+        a = proto.Client(sock_dir=sock_dir, pod_name="a").connect()
+        a.register()
+        b = proto.Client(sock_dir=sock_dir, pod_name="b").connect()
+        b.register()
+        a.send(proto.REQ_LOCK)
+        assert a.recv(5).type == proto.LOCK_OK
+        b.send(proto.REQ_LOCK)
+        assert a.recv(5).type == proto.DROP_LOCK
+        a.send(proto.LOCK_RELEASED)
+        assert b.recv(5).type == proto.LOCK_OK
+        a.close()  # eviction path
+        b.send(proto.LOCK_RELEASED)
+        b.close()
+        time.sleep(0.5)
+    finally:
+        p.send_signal(signal.SIGTERM)
+        try:
+            rc = p.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            rc = p.wait(timeout=5)
+        log.close()
+
+    text = Path(sock_dir, "tsan.log").read_text(errors="replace")
+    assert "WARNING: ThreadSanitizer" not in text, text[-4000:]
+    assert rc != 66, "TSan reported races"
