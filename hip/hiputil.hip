@@ -153,6 +153,51 @@ int nvs_hip_advise(const void *p, size_t bytes, int advice, int device)
 	return 0;
 }
 
+/* Chunked prefetch across n_streams concurrent streams (probes SDMA
+ * parallelism for post-preemption working-set restore). */
+int nvs_hip_prefetch_chunked(const void *p, size_t bytes, int device,
+			     int n_streams, size_t chunk_bytes)
+{
+	hipStream_t streams[16];
+	int i, n = n_streams;
+
+	if (n < 1)
+		n = 1;
+	if (n > 16)
+		n = 16;
+	if (chunk_bytes == 0)
+		chunk_bytes = 256ULL << 20;
+	for (i = 0; i < n; i++)
+		CHECK(hipStreamCreateWithFlags(&streams[i],
+					       hipStreamNonBlocking));
+	{
+		const char *base = (const char *)p;
+		size_t off = 0;
+		int s = 0;
+
+		while (off < bytes) {
+			size_t len = bytes - off < chunk_bytes ?
+				     bytes - off : chunk_bytes;
+			hipError_t e = hipMemPrefetchAsync(
+				base + off, len, device, streams[s]);
+			if (e != hipSuccess) {
+				for (i = 0; i < n; i++)
+					hipStreamDestroy(streams[i]);
+				return (int)e;
+			}
+			off += len;
+			s = (s + 1) % n;
+		}
+	}
+	for (i = 0; i < n; i++) {
+		hipError_t e = hipStreamSynchronize(streams[i]);
+		hipStreamDestroy(streams[i]);
+		if (e != hipSuccess)
+			return (int)e;
+	}
+	return 0;
+}
+
 int nvs_hip_sync(void)
 {
 	CHECK(hipDeviceSynchronize());

@@ -48,6 +48,14 @@ class HipUtil:
     ADVISE_ACCESSED_BY = 5
     ADVISE_COARSE_GRAIN = 100
 
+    def prefetch_chunked(self, ptr: int, nbytes: int, device: int = 0,
+                         n_streams: int = 4,
+                         chunk_bytes: int = 256 << 20) -> None:
+        self._ck(self.lib.nvs_hip_prefetch_chunked(
+            ctypes.c_void_p(ptr), ctypes.c_size_t(nbytes), device,
+            n_streams, ctypes.c_size_t(chunk_bytes)),
+            "prefetch_chunked")
+
     def advise(self, ptr: int, nbytes: int, advice: int,
                device: int = 0) -> None:
         self._ck(self.lib.nvs_hip_advise(ctypes.c_void_p(ptr),
@@ -92,6 +100,7 @@ def load() -> HipUtil:
     lib = ctypes.CDLL(str(path))
     for fname in ("nvs_hip_malloc_managed", "nvs_hip_malloc",
                   "nvs_hip_free", "nvs_hip_prefetch", "nvs_hip_advise",
+                  "nvs_hip_prefetch_chunked",
                   "nvs_touch_pages", "nvs_read_pages", "nvs_busy",
                   "nvs_stream_triad", "nvs_hip_sync",
                   "nvs_hip_mem_get_info"):
