@@ -24,6 +24,7 @@ def client_env(
     reserve_mib: int | None = None,
     fake_total_mib: int | None = None,
     prefetch: bool | None = None,
+    evict: bool | None = None,
     disable_um: bool = False,
     use_stub: bool = False,
     extra: Mapping[str, str] | None = None,
@@ -50,6 +51,8 @@ def client_env(
         env["NVSHARE_FAKE_TOTAL_MIB"] = str(fake_total_mib)
     if prefetch is not None:
         env["NVSHARE_PREFETCH"] = "1" if prefetch else "0"
+    if evict is not None:
+        env["NVSHARE_EVICT"] = "1" if evict else "0"
     if disable_um:
         env["NVSHARE_DISABLE_UM"] = "1"
 

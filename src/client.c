@@ -254,6 +254,8 @@ static void handle_drop_lock(void)
 		return; /* already released voluntarily */
 	drain_gpu();
 	send_msg_type(NVS_LOCK_RELEASED);
+	if (nvs_env_bool("NVSHARE_EVICT", 0))
+		nvs_evict_allocs();
 	log_debug("client: lock released after DROP_LOCK");
 }
 

@@ -29,6 +29,8 @@ struct nvs_real_hip {
 	fn_hipSetDevice hipSetDevice;
 	fn_hipGetDevice hipGetDevice;
 	fn_hipStreamSynchronize hipStreamSynchronize;
+	fn_hipStreamCreateWithFlags hipStreamCreateWithFlags;
+	fn_hipStreamDestroy hipStreamDestroy;
 	fn_hipLaunchKernel hipLaunchKernel;
 	fn_hipExtLaunchKernel hipExtLaunchKernel;
 	fn_hipLaunchCooperativeKernel hipLaunchCooperativeKernel;
@@ -79,5 +81,9 @@ int nvs_scheduler_gating(void);
 /* Managed-allocation registry hooks (hook.c) used for prefetch. Called
  * with the list snapshot under the allocation lock. */
 void nvs_prefetch_allocs(void);
+
+/* Migrate the tracked working set to host after a lock release
+ * (NVSHARE_EVICT=1). */
+void nvs_evict_allocs(void);
 
 #endif /* NVSHARE_CLIENT_H */

@@ -64,6 +64,10 @@ typedef nvshipError_t (*fn_hipDeviceSynchronize)(void);
 typedef nvshipError_t (*fn_hipSetDevice)(int);
 typedef nvshipError_t (*fn_hipGetDevice)(int *);
 typedef nvshipError_t (*fn_hipStreamSynchronize)(nvship_stream_t);
+typedef nvshipError_t (*fn_hipStreamCreateWithFlags)(nvship_stream_t *,
+						     unsigned int);
+typedef nvshipError_t (*fn_hipStreamDestroy)(nvship_stream_t);
+#define NVSHIP_STREAM_NON_BLOCKING 0x01
 typedef nvshipError_t (*fn_hipLaunchKernel)(const void *, nvship_dim3,
 					    nvship_dim3, void **, size_t,
 					    nvship_stream_t);

@@ -188,6 +188,20 @@ hipError_t hipStreamSynchronize(hipStream_t s)
 	return 0;
 }
 
+hipError_t hipStreamCreateWithFlags(hipStream_t *s, unsigned int flags)
+{
+	(void)flags;
+	if (s)
+		*s = (hipStream_t)0x1;
+	return 0;
+}
+
+hipError_t hipStreamDestroy(hipStream_t s)
+{
+	(void)s;
+	return 0;
+}
+
 static hipError_t run_kernel(void)
 {
 	stub_init();
