@@ -20,17 +20,27 @@ def main(argv: list[str] | None = None) -> None:
     ap.add_argument("--gb", type=float, default=12.0,
                     help="approx total working set in GiB (two tensors)")
     ap.add_argument("--iters", type=int, default=1000)
+    ap.add_argument("--think-every", type=int, default=0,
+                    help="sleep --think-s after every N iters (emulates "
+                    "the reference's GPU/CPU-mixed *_50 workloads, "
+                    "thesis Table 12.1)")
+    ap.add_argument("--think-s", type=float, default=2.0)
     args = ap.parse_args(argv)
 
     import torch
+
+    import time
 
     n = int(math.sqrt(args.gb * (1 << 30) / 2 / 4))
     dev = torch.device(args.device)
     with Timer() as t:
         x = torch.ones((n, n), dtype=torch.float32, device=dev)
         y = torch.zeros((n, n), dtype=torch.float32, device=dev)
-        for _ in range(args.iters):
+        for i in range(args.iters):
             y.add_(x)
+            if args.think_every and (i + 1) % args.think_every == 0:
+                sync(args.device)
+                time.sleep(args.think_s)
         sync(args.device)
         expect = float(args.iters)
         got = y[n // 2, n // 2].item()

@@ -18,6 +18,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import subprocess
 import sys
 import tempfile
 import time
@@ -40,11 +41,23 @@ def main() -> None:
     ap.add_argument("--gb", type=float, default=0.0,
                     help="per-job working set (default: mode-derived)")
     ap.add_argument("--iters", type=int, default=300)
+    ap.add_argument("--think-every", type=int, default=0,
+                    help="pytorch_add think phases (emulate *_50 "
+                         "GPU/CPU-mixed jobs)")
+    ap.add_argument("--think-s", type=float, default=2.0)
     ap.add_argument("--tqs", default="30,10,2",
                     help="comma list of TQ values to sweep")
     ap.add_argument("--jobs", type=int, default=2)
     ap.add_argument("--skip-off", action="store_true",
                     help="skip the scheduler-off (thrashing) arm")
+    ap.add_argument("--off-timeout", type=float, default=0.0,
+                    help="separate (shorter) timeout for the "
+                         "scheduler-off thrash arm")
+    ap.add_argument("--squat-leave-gb", type=float, default=0.0,
+                    help="pin HBM with tools/squatter.py leaving this "
+                         "many GB free (REAL oversubscription pressure)")
+    ap.add_argument("--migration-assist", action="store_true",
+                    help="enable NVSHARE_PREFETCH + NVSHARE_EVICT")
     ap.add_argument("--timeout", type=float, default=1800.0)
     ap.add_argument("--out", default="profiles/colo.json")
     args = ap.parse_args()
@@ -59,16 +72,43 @@ def main() -> None:
 
     cmd = workload_cmd("pytorch_add", "--gb", str(args.gb), "--iters",
                        str(args.iters))
+    if args.think_every:
+        cmd += ["--think-every", str(args.think_every), "--think-s",
+                str(args.think_s)]
     env_kwargs = {}
     if args.fake_total_mib:
         env_kwargs = {"fake_total_mib": args.fake_total_mib,
                       "oversubscribe": True,
                       "reserve_mib": max(64, args.fake_total_mib // 32)}
+    if args.squat_leave_gb > 0:
+        # Real pressure: clients are capped by *advertised* capacity
+        # which no longer matches free HBM, so let them oversubscribe.
+        env_kwargs.setdefault("oversubscribe", True)
+    if args.migration_assist:
+        env_kwargs["prefetch"] = True
+        env_kwargs["evict"] = True
+    if args.think_every:
+        # Idle probes must be finer than the think phases for early
+        # release to overlap one job's CPU time with the other's GPU.
+        env_kwargs["extra"] = {"NVSHARE_RELEASE_INTERVAL_S": "1"}
+
+    squat = None
+    if args.squat_leave_gb > 0:
+        squat = subprocess.Popen(
+            [sys.executable, str(REPO / "tools" / "squatter.py"),
+             "--leave-gb", str(args.squat_leave_gb),
+             "--seconds", "86400"],
+            stdout=subprocess.PIPE, text=True)
+        line = squat.stdout.readline()
+        assert "SQUATTING" in line, line
+        print(line.strip(), flush=True)
 
     sock_dir = tempfile.mkdtemp(prefix="nvs-tq-", dir="/tmp")
     out: dict = {"mode": args.mode, "gb_per_job": args.gb,
                  "iters": args.iters, "jobs": args.jobs,
-                 "fake_total_mib": args.fake_total_mib, "rows": {}}
+                 "fake_total_mib": args.fake_total_mib,
+                 "squat_leave_gb": args.squat_leave_gb,
+                 "migration_assist": args.migration_assist, "rows": {}}
 
     with SchedulerDaemon(sock_dir=sock_dir, tq=30):
         # solo + serial baseline (scheduler on, but only one client)
@@ -96,13 +136,17 @@ def main() -> None:
             time.sleep(0.5)
             r = run_colocated([cmd] * args.jobs, sock_dir=sock_dir,
                               env_kwargs=env_kwargs,
-                              timeout=args.timeout)
+                              timeout=args.off_timeout or args.timeout)
             out["rows"]["parallel_sched_off"] = (
                 r.makespan if r.ok else None)
+            out["rows"]["parallel_sched_off_dnf"] = not r.ok
             ctl.set_scheduling(True, sock_dir)
             print(f"sched-off: "
-                  f"{'%.1fs' % r.makespan if r.ok else 'FAILED'}",
+                  f"{'%.1fs' % r.makespan if r.ok else 'TIMEOUT/DNF'}",
                   flush=True)
+
+    if squat is not None:
+        squat.terminate()
 
     o = Path(args.out)
     o.parent.mkdir(parents=True, exist_ok=True)
