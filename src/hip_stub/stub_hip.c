@@ -30,6 +30,7 @@ typedef struct { unsigned x, y, z; } dim3_t;
 static pthread_mutex_t log_mutex = PTHREAD_MUTEX_INITIALIZER;
 static int log_fd = -2; /* -2 = uninitialized, -1 = disabled */
 static long kernel_us;
+static long sync_us;
 static size_t total_bytes;
 static size_t used_bytes;
 static pthread_mutex_t mem_mutex = PTHREAD_MUTEX_INITIALIZER;
@@ -54,6 +55,8 @@ static void stub_init(void)
 			open(v, O_WRONLY | O_CREAT | O_APPEND, 0666) : -1;
 		v = getenv("NVSTUB_KERNEL_US");
 		kernel_us = v != NULL ? atol(v) : 0;
+		v = getenv("NVSTUB_SYNC_US");
+		sync_us = v != NULL ? atol(v) : 0;
 		v = getenv("NVSTUB_TOTAL_MIB");
 		total_bytes = (v != NULL ? (size_t)atol(v) : 1024)
 			* 1024 * 1024;
@@ -164,7 +167,10 @@ hipError_t hipMemAdvise(const void *p, size_t n, int advice, int dev)
 
 hipError_t hipDeviceSynchronize(void)
 {
+	stub_init();
 	ev("hipDeviceSynchronize", 0);
+	if (sync_us > 0)
+		usleep((useconds_t)sync_us);
 	return 0;
 }
 

@@ -161,6 +161,34 @@ def test_two_clients_serialized(artifacts, sched, sock_dir):
     assert "DROP_LOCK" in sched_log
 
 
+def test_pending_window_adapts(artifacts, sched, sock_dir):
+    """Slow drains shrink the pending-kernel window, fast drains grow
+    it (reference hook.c:782-838 behavior)."""
+    # Slow syncs (1.2 s) => window collapses toward 1.
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      8, "--iters", 40,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024",
+                                "NVSTUB_SYNC_US": "1200000"},
+                      env_extra={"NVSHARE_WINDOW_START": "4"},
+                      reserve_mib=64, timeout=120)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    lines = [l for l in r.stderr.splitlines() if "window sync" in l]
+    assert lines, r.stderr
+    # last adaptation must have shrunk the window to the minimum
+    assert "window=1" in lines[-1] or "window=2" in lines[-1], lines
+
+    # Fast syncs => window doubles up to the max.
+    r = run_hipclient(artifacts, sock_dir, "--allocs", 1, "--alloc-mib",
+                      8, "--iters", 300,
+                      stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                      env_extra={"NVSHARE_WINDOW_START": "4",
+                                 "NVSHARE_WINDOW_MAX": "64"},
+                      reserve_mib=64)
+    assert r.returncode == 0
+    lines = [l for l in r.stderr.splitlines() if "window sync" in l]
+    assert lines and "window=64" in lines[-1], lines[-3:]
+
+
 def test_scheduler_restart_reconnect(artifacts, sock_dir):
     """Clients survive a scheduler restart (reference killed the app)."""
     from nvshare_amd.scheduler import SchedulerDaemon
