@@ -73,6 +73,14 @@ def reexec_under_preload(args, device: str) -> None:
     rank = int(os.environ.get("RANK", "0"))
     sock_dir = os.environ.get("NVSHARE_BENCH_SOCK_DIR",
                               "/tmp/nvshare-bench")
+    spath = proto.scheduler_path(sock_dir)
+    if rank == 0 and os.path.exists(spath):
+        # Stale socket from a dead daemon? Probe it.
+        try:
+            from nvshare_amd import ctl
+            ctl.status(sock_dir, timeout=2.0)
+        except Exception:
+            os.unlink(spath)
     if rank == 0 and not os.path.exists(proto.scheduler_path(sock_dir)):
         # Daemonize the scheduler for the duration of the bench; it is
         # torn down by the last rank (best effort) or just left idle.
@@ -96,6 +104,9 @@ def reexec_under_preload(args, device: str) -> None:
         fake_total_mib=args.oversub_fake_mib or None,
         extra={
             PRELOAD_GUARD: "1",
+            # deterministic MIOpen warmup on fresh boxes
+            "MIOPEN_FIND_MODE": os.environ.get("MIOPEN_FIND_MODE",
+                                               "FAST"),
             # all ranks co-locate on physical GPU 0
             "HIP_VISIBLE_DEVICES": os.environ.get(
                 "NVSHARE_BENCH_GPU", "0"),
