@@ -233,13 +233,15 @@ static int connect_and_register(void)
 
 static void handle_lock_ok(void)
 {
+	/* Restore the working set BEFORE opening the gate: demand
+	 * faults racing the bulk migration degrade both. */
+	if (nvs_env_bool("NVSHARE_PREFETCH", 0))
+		nvs_prefetch_allocs();
 	pthread_mutex_lock(&g_mutex);
 	own_lock = 1;
 	need_lock = 0;
 	pthread_cond_broadcast(&own_lock_cv);
 	pthread_mutex_unlock(&g_mutex);
-	if (nvs_env_bool("NVSHARE_PREFETCH", 0))
-		nvs_prefetch_allocs();
 }
 
 static void handle_drop_lock(void)
@@ -253,9 +255,9 @@ static void handle_drop_lock(void)
 	if (!had)
 		return; /* already released voluntarily */
 	drain_gpu();
-	send_msg_type(NVS_LOCK_RELEASED);
 	if (nvs_env_bool("NVSHARE_EVICT", 0))
-		nvs_evict_allocs();
+		nvs_evict_allocs(); /* blocking; under pressure only */
+	send_msg_type(NVS_LOCK_RELEASED);
 	log_debug("client: lock released after DROP_LOCK");
 }
 

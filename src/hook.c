@@ -374,11 +374,33 @@ static void ensure_prefetch_streams(void)
 	prefetch_streams_ready = 1;
 }
 
+static void sync_prefetch_streams(void)
+{
+	int i;
+
+	if (real.hipStreamSynchronize == NULL)
+		return;
+	for (i = 0; i < PREFETCH_STREAMS; i++)
+		real.hipStreamSynchronize(prefetch_streams[i]);
+}
+
+/* Real (driver) free memory — NOT the advertised lie. */
+static size_t real_free_bytes(void)
+{
+	size_t freeb = 0, totalb = 0;
+
+	if (real.hipMemGetInfo == NULL ||
+	    real.hipMemGetInfo(&freeb, &totalb) != NVSHIP_SUCCESS)
+		return (size_t)-1;
+	return freeb;
+}
+
 void nvs_prefetch_allocs(void)
 {
 	struct nvs_alloc *a;
 	size_t budget;
 	int s = 0;
+	int64_t t0 = nvs_now_ns();
 
 	if (real.hipMemPrefetchAsync == NULL)
 		return;
@@ -406,6 +428,12 @@ void nvs_prefetch_allocs(void)
 		}
 	}
 	pthread_mutex_unlock(&alloc_mutex);
+	/* Blocking: the gate opens only after the working set is back,
+	 * otherwise the app's demand faults (0.15 GB/s) race the bulk
+	 * migration (3.8 GB/s) and lose. */
+	sync_prefetch_streams();
+	log_debug("hook: restore prefetch took %lld ms",
+		  (long long)((nvs_now_ns() - t0) / 1000000));
 }
 
 /*
@@ -420,10 +448,24 @@ void nvs_evict_allocs(void)
 {
 	struct nvs_alloc *a;
 	int s = 0;
+	size_t resident, freeb;
+	int64_t t0 = nvs_now_ns();
 
 	if (real.hipMemPrefetchAsync == NULL)
 		return;
 	ensure_prefetch_streams();
+	/* Only evict under actual memory pressure: when the device has
+	 * room for another client's set alongside ours, eviction would
+	 * just churn migrations. */
+	pthread_mutex_lock(&alloc_mutex);
+	resident = sum_allocated;
+	pthread_mutex_unlock(&alloc_mutex);
+	freeb = real_free_bytes();
+	if (freeb != (size_t)-1 && freeb > resident) {
+		log_debug("hook: skip evict (free %zu MiB > set %zu MiB)",
+			  freeb / NVS_MIB, resident / NVS_MIB);
+		return;
+	}
 	pthread_mutex_lock(&alloc_mutex);
 	for (a = alloc_list; a != NULL; a = a->next) {
 		char *p = a->ptr;
@@ -442,6 +484,10 @@ void nvs_evict_allocs(void)
 		}
 	}
 	pthread_mutex_unlock(&alloc_mutex);
+	/* Blocking: LOCK_RELEASED is sent after the room exists. */
+	sync_prefetch_streams();
+	log_debug("hook: evict to host took %lld ms",
+		  (long long)((nvs_now_ns() - t0) / 1000000));
 }
 
 /* ------------------------------------------------------------------ */
