@@ -426,12 +426,43 @@ static void *early_release_thread(void *arg)
 	return NULL;
 }
 
+/*
+ * fork() safety: a forked child (e.g. a DataLoader worker) inherits
+ * the preload state but not our threads, and shares the parent's
+ * scheduler socket fd — writing to it would corrupt the parent's
+ * protocol stream.  The child therefore detaches: fresh (unlocked)
+ * synchronization objects and free-running standalone mode.  GPU work
+ * from forked children is NOT arbitrated (same practical stance as the
+ * reference, which would deadlock instead); CUDA-style spawn children
+ * bootstrap their own client as usual.
+ */
+static void atfork_child(void)
+{
+	pthread_mutex_t fresh_m = PTHREAD_MUTEX_INITIALIZER;
+	pthread_cond_t fresh_c = PTHREAD_COND_INITIALIZER;
+	pthread_rwlock_t fresh_rw = PTHREAD_RWLOCK_INITIALIZER;
+
+	memcpy(&g_mutex, &fresh_m, sizeof(fresh_m));
+	memcpy(&sock_mutex, &fresh_m, sizeof(fresh_m));
+	memcpy(&own_lock_cv, &fresh_c, sizeof(fresh_c));
+	memcpy(&submit_rwlock, &fresh_rw, sizeof(fresh_rw));
+	if (sock_fd >= 0)
+		close(sock_fd);
+	sock_fd = -1;
+	standalone = 1;
+	scheduler_on = 0;
+	own_lock = 0;
+	need_lock = 0;
+}
+
 void nvs_client_init(void)
 {
 	pthread_t tid;
 	long timeout_s;
 	double deadline;
 	int fd = -1;
+
+	pthread_atfork(NULL, NULL, atfork_child);
 
 	release_interval_s = nvs_env_long("NVSHARE_RELEASE_INTERVAL_S", 5,
 					  1, 3600);

@@ -161,6 +161,22 @@ def test_two_clients_serialized(artifacts, sched, sock_dir):
     assert "DROP_LOCK" in sched_log
 
 
+def test_fork_safety(artifacts, sched, sock_dir):
+    """A forked child free-runs without corrupting the parent's
+    scheduler protocol or deadlocking (reference would deadlock)."""
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     reserve_mib=64)
+    env["NVSTUB_TOTAL_MIB"] = "1024"
+    r = subprocess.run([str(artifacts.hipclient.parent / "forkclient")],
+                       env=env, capture_output=True, text=True,
+                       timeout=60)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "PASS" in r.stdout
+    # Parent must still be a functioning client afterwards.
+    log = sched.log_text()
+    assert log.count("registered client") == 1
+
+
 def test_pending_window_adapts(artifacts, sched, sock_dir):
     """Slow drains shrink the pending-kernel window, fast drains grow
     it (reference hook.c:782-838 behavior)."""
