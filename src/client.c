@@ -116,6 +116,9 @@ void nvs_submit_begin(void)
 		if (!scheduler_on || own_lock) {
 			did_work = 1;
 			pthread_mutex_unlock(&g_mutex);
+			/* Deferred managed-page population happens here,
+			 * under the lock, before the real submission. */
+			nvs_populate_pending();
 			return; /* read lock held */
 		}
 		pthread_rwlock_unlock(&submit_rwlock);
@@ -145,6 +148,11 @@ void nvs_submit_end(void)
 int nvs_scheduler_gating(void)
 {
 	return scheduler_on && !standalone;
+}
+
+int nvs_can_submit_now(void)
+{
+	return standalone || !scheduler_on || own_lock;
 }
 
 /* Drain all outstanding GPU work (caller must NOT hold g_mutex). */

@@ -78,6 +78,13 @@ extern int nvs_app_device;
 /* 1 while a scheduler is actively gating this process. */
 int nvs_scheduler_gating(void);
 
+/* 1 when this process may submit GPU work right now (racy read). */
+int nvs_can_submit_now(void);
+
+/* Populate managed allocations whose eager prefetch was deferred
+ * because we didn't hold the GPU lock at hipMalloc time (hook.c). */
+void nvs_populate_pending(void);
+
 /* Managed-allocation registry hooks (hook.c) used for prefetch. Called
  * with the list snapshot under the allocation lock. */
 void nvs_prefetch_allocs(void);

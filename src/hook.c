@@ -82,10 +82,12 @@ static void *real_dlsym(void *handle, const char *name)
 struct nvs_alloc {
 	void *ptr;
 	size_t size;
+	int populated;   /* device pages materialized (eager prefetch) */
 	struct nvs_alloc *next;
 };
 static struct nvs_alloc *alloc_list;
 static size_t sum_allocated;
+static long n_unpopulated; /* atomic; fast-path check in the gate */
 static pthread_mutex_t alloc_mutex = PTHREAD_MUTEX_INITIALIZER;
 
 static size_t mem_total;        /* advertised total (bytes) */
@@ -285,20 +287,33 @@ static size_t mem_limit(void)
  * touches at HBM speed.  So every converted allocation is (a) advised
  * coarse-grain (whole-range migration granularity, full-rate access;
  * fine-grain host/device *concurrent* access is not something the
- * hipMalloc contract we replace ever promised) and (b) eagerly
- * prefetched to the device so pages are born resident instead of
- * being demand-faulted one 4 KiB page at a time.
+ * hipMalloc contract we replace ever promised) and (b) prefetched to
+ * the device so pages are born resident instead of demand-faulted one
+ * 4 KiB page at a time.
+ *
+ * The prefetch is GPU (SDMA) work and therefore gated: a co-located
+ * client populating 6 GB mid-quantum forcibly evicts the lock
+ * holder's resident pages and the two migration flows livelock at
+ * fault speed (observed: a 26 s experiment became a >1450 s hang).
+ * If we hold the lock at hipMalloc time, populate immediately;
+ * otherwise the range is marked pending and populated at the first
+ * gated submission (nvs_populate_pending, called under the lock).
+ * Returns 1 when populated now.
  */
-static void populate_managed(void *ptr, size_t size)
+static int populate_managed(void *ptr, size_t size)
 {
 	if (coarse_grain && real.hipMemAdvise != NULL)
 		real.hipMemAdvise(ptr, size,
 				  NVSHIP_MEM_ADVISE_SET_COARSE_GRAIN, 0);
-	if (alloc_prefetch && real.hipMemPrefetchAsync != NULL)
-		real.hipMemPrefetchAsync(ptr, size, nvs_app_device, NULL);
+	if (!alloc_prefetch || real.hipMemPrefetchAsync == NULL)
+		return 1; /* nothing to defer */
+	if (!nvs_can_submit_now())
+		return 0; /* defer to the gate */
+	real.hipMemPrefetchAsync(ptr, size, nvs_app_device, NULL);
+	return 1;
 }
 
-static void track_alloc(void *ptr, size_t size)
+static void track_alloc(void *ptr, size_t size, int populated)
 {
 	struct nvs_alloc *a = malloc(sizeof(*a));
 
@@ -306,13 +321,43 @@ static void track_alloc(void *ptr, size_t size)
 		return;
 	a->ptr = ptr;
 	a->size = size;
+	a->populated = populated;
 	pthread_mutex_lock(&alloc_mutex);
 	a->next = alloc_list;
 	alloc_list = a;
 	sum_allocated += size;
+	if (!populated)
+		__atomic_fetch_add(&n_unpopulated, 1, __ATOMIC_RELAXED);
 	pthread_mutex_unlock(&alloc_mutex);
-	log_debug("hook: +alloc %p %zu MiB (sum %zu MiB)", ptr,
-		  size / NVS_MIB, sum_allocated / NVS_MIB);
+	log_debug("hook: +alloc %p %zu MiB (sum %zu MiB%s)", ptr,
+		  size / NVS_MIB, sum_allocated / NVS_MIB,
+		  populated ? "" : ", population deferred");
+}
+
+/* Called from the gate with the submission read lock held and the GPU
+ * lock owned: materialize any deferred ranges before real work. */
+void nvs_populate_pending(void)
+{
+	struct nvs_alloc *a;
+	int any = 0;
+
+	if (__atomic_load_n(&n_unpopulated, __ATOMIC_RELAXED) == 0)
+		return;
+	if (real.hipMemPrefetchAsync == NULL)
+		return;
+	pthread_mutex_lock(&alloc_mutex);
+	for (a = alloc_list; a != NULL; a = a->next) {
+		if (a->populated)
+			continue;
+		real.hipMemPrefetchAsync(a->ptr, a->size, nvs_app_device,
+					 NULL);
+		a->populated = 1;
+		__atomic_fetch_sub(&n_unpopulated, 1, __ATOMIC_RELAXED);
+		any = 1;
+	}
+	pthread_mutex_unlock(&alloc_mutex);
+	if (any)
+		log_debug("hook: populated deferred ranges");
 }
 
 /* Returns tracked size, or 0 if unknown pointer. */
@@ -328,6 +373,9 @@ static size_t untrack_alloc(void *ptr)
 			*pp = a->next;
 			size = a->size;
 			sum_allocated -= size;
+			if (!a->populated)
+				__atomic_fetch_sub(&n_unpopulated, 1,
+						   __ATOMIC_RELAXED);
 			free(a);
 			break;
 		}
@@ -414,6 +462,12 @@ void nvs_prefetch_allocs(void)
 	for (a = alloc_list; a != NULL && budget > 0; a = a->next) {
 		size_t left = a->size < budget ? a->size : budget;
 		char *p = a->ptr;
+
+		if (!a->populated) {
+			a->populated = 1;
+			__atomic_fetch_sub(&n_unpopulated, 1,
+					   __ATOMIC_RELAXED);
+		}
 
 		while (left > 0) {
 			size_t n = left < PREFETCH_CHUNK ? left :
@@ -628,10 +682,8 @@ nvshipError_t hipMalloc(void **ptr, size_t size)
 		pthread_mutex_unlock(&alloc_mutex);
 	}
 	r = real.hipMallocManaged(ptr, size, NVSHIP_MEM_ATTACH_GLOBAL);
-	if (r == NVSHIP_SUCCESS) {
-		populate_managed(*ptr, size);
-		track_alloc(*ptr, size);
-	}
+	if (r == NVSHIP_SUCCESS)
+		track_alloc(*ptr, size, populate_managed(*ptr, size));
 	return r;
 }
 
@@ -1017,7 +1069,7 @@ nvshipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags)
 	}
 	r = real.hipMallocManaged(ptr, size, flags);
 	if (r == NVSHIP_SUCCESS && size > 0 && !disable_um)
-		track_alloc(*ptr, size);
+		track_alloc(*ptr, size, 1); /* app-managed: don't touch */
 	return r;
 }
 

@@ -58,6 +58,7 @@ struct conn {
 	char pod_name[NVS_POD_NAME_LEN];
 	char pod_namespace[NVS_POD_NS_LEN];
 	int wants_lock;           /* present in its GPU's request queue */
+	int relock_pending;       /* holder re-requested during release */
 	struct conn *next;        /* registry list */
 	struct conn *qnext;       /* FCFS queue list */
 };
@@ -356,7 +357,14 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 		}
 		if (!scheduler_on)
 			break; /* clients free-run while scheduling is off */
-		if (c->wants_lock || g->lock_holder == c)
+		if (g->lock_holder == c) {
+			/* The holder re-requests between dropping
+			 * own_lock and LOCK_RELEASED arriving: queue it
+			 * again once the release is processed. */
+			c->relock_pending = 1;
+			break;
+		}
+		if (c->wants_lock)
 			break;
 		queue_push(g, c);
 		if (!g->lock_held)
@@ -376,6 +384,10 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 		g->lock_held = 0;
 		g->lock_holder = NULL;
 		g->drop_lock_sent = 0;
+		if (c->relock_pending) {
+			c->relock_pending = 0;
+			queue_push(g, c);
+		}
 		try_schedule(g);
 		break;
 	case NVS_SET_TQ: {
