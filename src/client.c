@@ -123,6 +123,8 @@ void nvs_submit_begin(void)
 		}
 		pthread_rwlock_unlock(&submit_rwlock);
 		for (;;) {
+			struct timespec abs;
+
 			if (!scheduler_on || own_lock)
 				break;
 			if (!need_lock) {
@@ -132,7 +134,15 @@ void nvs_submit_begin(void)
 				if (send_msg_type(NVS_REQ_LOCK) != 0)
 					need_lock = 0; /* retry next pass */
 			}
-			pthread_cond_wait(&own_lock_cv, &g_mutex);
+			/* Re-send REQ_LOCK if nothing arrives for 10 s:
+			 * idempotent at the scheduler, and insurance
+			 * against lost-wakeup protocol bugs. */
+			clock_gettime(CLOCK_REALTIME, &abs);
+			abs.tv_sec += 10;
+			if (pthread_cond_timedwait(&own_lock_cv, &g_mutex,
+						   &abs) != 0 &&
+			    scheduler_on && !own_lock)
+				need_lock = 0;
 		}
 		pthread_mutex_unlock(&g_mutex);
 	}
