@@ -76,7 +76,26 @@ __global__ void stream_triad_kernel(float *__restrict__ a,
 	}
 }
 
+__global__ void tiny_kernel(float *p)
+{
+	if (threadIdx.x == 0 && blockIdx.x == 0)
+		p[0] += 1.0f;
+}
+
 extern "C" {
+
+/* Launch n trivial kernels touching p; measures per-launch overhead
+ * (managed vs plain pointers — isolates the launch-dense workload
+ * penalty of the managed conversion). */
+int nvs_launch_burst(float *p, int n)
+{
+	for (int i = 0; i < n; i++)
+		hipLaunchKernelGGL(tiny_kernel, dim3(1), dim3(64), 0,
+				   nullptr, p);
+	CHECK(hipGetLastError());
+	CHECK(hipDeviceSynchronize());
+	return 0;
+}
 
 /* Touch every `stride`-th float of buf[0..n) from the GPU. */
 int nvs_touch_pages(float *buf, size_t n, size_t stride, float val,
