@@ -110,7 +110,7 @@ def reexec_under_preload(args, device: str) -> None:
             # all ranks co-locate on physical GPU 0
             "HIP_VISIBLE_DEVICES": os.environ.get(
                 "NVSHARE_BENCH_GPU", "0"),
-            "NVSHARE_RELEASE_INTERVAL_S": "1",
+            "NVSHARE_RELEASE_INTERVAL_MS": "200",
             "NVSHARE_POD_NAME": f"bench-rank{rank}",
         },
     )
@@ -156,7 +156,9 @@ def main(argv=None):
     if device == "cuda":
         torch.backends.cudnn.benchmark = True  # MIOpen autotune
         model = model.to(memory_format=torch.channels_last)
-    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    # lr low enough that bf16 training on random labels stays finite
+    # over any K the driver picks.
+    opt = torch.optim.SGD(model.parameters(), lr=0.02, momentum=0.9)
     lossf = torch.nn.CrossEntropyLoss()
     x = torch.randn(batch, 3, image, image, device=dev)
     if device == "cuda":

@@ -89,7 +89,7 @@ static fn_rsmi_init p_rsmi_init;
 static fn_rsmi_dev_busy_percent_get p_rsmi_busy;
 static int rsmi_ready = 0;
 static uint32_t rsmi_dev_index = 0;
-static long release_interval_s = 5;
+static long release_interval_ms = 5000;
 static long idle_sync_threshold_ms = 100;
 
 static int send_msg_type(uint8_t type)
@@ -419,7 +419,7 @@ static void *early_release_thread(void *arg)
 	for (;;) {
 		int worked, have_lock;
 
-		usleep((useconds_t)(release_interval_s * 1000000));
+		usleep((useconds_t)(release_interval_ms * 1000));
 		pthread_mutex_lock(&g_mutex);
 		have_lock = own_lock && scheduler_on;
 		worked = did_work;
@@ -482,8 +482,14 @@ void nvs_client_init(void)
 
 	pthread_atfork(NULL, NULL, atfork_child);
 
-	release_interval_s = nvs_env_long("NVSHARE_RELEASE_INTERVAL_S", 5,
-					  1, 3600);
+	/* Millisecond-granularity idle probing: bursty/barrier-waiting
+	 * clients hand the GPU over in ~2x this interval.  The coarser
+	 * reference default was 5 s (client.c:51). */
+	release_interval_ms = nvs_env_long("NVSHARE_RELEASE_INTERVAL_MS",
+					   0, 0, 3600000);
+	if (release_interval_ms == 0)
+		release_interval_ms = nvs_env_long(
+			"NVSHARE_RELEASE_INTERVAL_S", 5, 1, 3600) * 1000;
 	idle_sync_threshold_ms = nvs_env_long("NVSHARE_IDLE_SYNC_MS", 100,
 					      1, 60000);
 	timeout_s = nvs_env_long("NVSHARE_CONNECT_TIMEOUT_S", 30, 1, 3600);

@@ -169,6 +169,43 @@ def test_hook_counters_dump(sched, sock_dir):
     assert "hipMalloc=" in r.stderr or "hipMalloc " in r.stderr
 
 
+def test_colocated_numerics_deterministic(sched, sock_dir):
+    """Two co-located clients computing a seeded matmul chain under
+    preemption must both match the CPU fp32 reference exactly —
+    catches any migration/corruption bug in the sharing path."""
+    code = (
+        "import torch\n"
+        "torch.manual_seed(7)\n"
+        "a = torch.randn(256, 256)\n"
+        "ref = a.clone()\n"
+        "for _ in range(30):\n"
+        "    ref = (ref @ a).clamp(-10, 10) / 10 + a\n"
+        "g = a.cuda()\n"
+        "x = a.cuda()\n"
+        "for _ in range(30):\n"
+        "    x = (x @ g).clamp(-10, 10) / 10 + g\n"
+        "err = (ref - x.cpu()).abs().max().item()\n"
+        "assert err < 1e-4, f'max err {err}'\n"
+        "print('NUMERICS_OK', err)\n"
+    )
+    import threading
+
+    results = []
+
+    def one():
+        r = run_torch_client(code, sock_dir, timeout=300)
+        results.append(r)
+
+    ts = [threading.Thread(target=one) for _ in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    for r in results:
+        assert r.returncode == 0, r.stderr[-2000:]
+        assert "NUMERICS_OK" in r.stdout
+
+
 def test_train_resnet50_loss_finite(sched, sock_dir):
     code = (
         "from nvshare_amd.workloads.train_resnet import run_training; "
