@@ -63,6 +63,8 @@ static pthread_rwlock_t submit_rwlock = PTHREAD_RWLOCK_INITIALIZER;
 static pthread_mutex_t sock_mutex = PTHREAD_MUTEX_INITIALIZER;
 static int sock_fd = -1;
 
+static volatile int shutting_down;
+
 static sem_t init_done_sem;
 static char pod_name[NVS_POD_NAME_LEN];
 static char pod_namespace[NVS_POD_NS_LEN];
@@ -169,11 +171,19 @@ int nvs_can_submit_now(void)
 static void drain_gpu(void)
 {
 	pthread_rwlock_wrlock(&submit_rwlock);
-	if (real.hipSetDevice != NULL)
+	if (!shutting_down && real.hipSetDevice != NULL)
 		real.hipSetDevice(nvs_app_device);
-	if (real.hipDeviceSynchronize != NULL)
+	if (!shutting_down && real.hipDeviceSynchronize != NULL)
 		real.hipDeviceSynchronize();
 	pthread_rwlock_unlock(&submit_rwlock);
+}
+
+/* exit() has begun: the HIP runtime / ROCm SMI may already be torn
+ * down, so the injected threads must stop touching them.  Registered
+ * with atexit (runs before library destructors). */
+static void on_process_exit(void)
+{
+	shutting_down = 1;
 }
 
 static void read_pod_identity(void)
@@ -388,6 +398,8 @@ static void idle_detect_init(void)
 /* 1 = GPU looks idle, 0 = busy/unknown. */
 static int check_idle(void)
 {
+	if (shutting_down)
+		return 0;
 	if (rsmi_ready) {
 		uint32_t busy = 100;
 
@@ -420,6 +432,8 @@ static void *early_release_thread(void *arg)
 		int worked, have_lock;
 
 		usleep((useconds_t)(release_interval_ms * 1000));
+		if (shutting_down)
+			continue; /* never touch HIP during teardown */
 		pthread_mutex_lock(&g_mutex);
 		have_lock = own_lock && scheduler_on;
 		worked = did_work;
@@ -481,6 +495,7 @@ void nvs_client_init(void)
 	int fd = -1;
 
 	pthread_atfork(NULL, NULL, atfork_child);
+	atexit(on_process_exit);
 
 	/* Millisecond-granularity idle probing: bursty/barrier-waiting
 	 * clients hand the GPU over in ~2x this interval.  The coarser
