@@ -62,6 +62,28 @@ def main() -> None:
         h, ctypes.c_void_p(p), args.n)
     h.free(p)
 
+    # Does the penalty scale with the NUMBER of live managed ranges?
+    # (a torch caching allocator holds dozens of segments)
+    for count in (16, 64, 256):
+        ptrs = []
+        for _ in range(count):
+            q = h.malloc_managed(1 << 20)
+            h.advise(q, 1 << 20, h.ADVISE_COARSE_GRAIN, 0)
+            h.prefetch(q, 1 << 20, 0)
+            ptrs.append(q)
+        h.sync()
+        res[f"managed_{count}_ranges_us"] = per_launch_us(
+            h, ctypes.c_void_p(ptrs[0]), args.n)
+        for q in ptrs:
+            h.free(q)
+
+    # Same count of PLAIN allocations for the control arm.
+    ptrs = [h.malloc(1 << 20) for _ in range(256)]
+    res["plain_256_ranges_us"] = per_launch_us(
+        h, ctypes.c_void_p(ptrs[0]), args.n)
+    for q in ptrs:
+        h.free(q)
+
     res["managed_penalty_us"] = (res["managed_coarse_prefetched_us"]
                                  - res["plain_us"])
     out = Path(args.out)
