@@ -18,6 +18,7 @@ class SchedulerStatus:
     tq_seconds: int
     clients: int
     queued: int
+    tracked_mib: int = 0
 
 
 def _one_shot(msg_type: int, data: str = "",
@@ -51,5 +52,7 @@ def status(sock_dir: str | None = None,
     reply = _one_shot(proto.STATUS_REQ, "", sock_dir, want_reply=True,
                       timeout=timeout)
     assert reply is not None and reply.type == proto.STATUS, reply
-    on, tq, ncl, qlen = (int(x) for x in reply.data.split(","))
-    return SchedulerStatus(bool(on), tq, ncl, qlen)
+    parts = [int(x) for x in reply.data.split(",")]
+    on, tq, ncl, qlen = parts[:4]
+    mib = parts[4] if len(parts) > 4 else 0
+    return SchedulerStatus(bool(on), tq, ncl, qlen, mib)

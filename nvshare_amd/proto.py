@@ -30,6 +30,7 @@ LOCK_RELEASED = 7
 SET_TQ = 8
 STATUS_REQ = 9
 STATUS = 10
+MEM_UPDATE = 11
 
 TYPE_NAMES = {
     REGISTER: "REGISTER",
@@ -42,6 +43,7 @@ TYPE_NAMES = {
     SET_TQ: "SET_TQ",
     STATUS_REQ: "STATUS_REQ",
     STATUS: "STATUS",
+    MEM_UPDATE: "MEM_UPDATE",
 }
 
 SOCK_DIR_DEFAULT = "/var/run/nvshare/"

@@ -428,12 +428,28 @@ static void *early_release_thread(void *arg)
 
 	idle_detect_init();
 
+	long last_mem_mib = -1;
+
 	for (;;) {
 		int worked, have_lock;
+		long mem_mib;
 
 		usleep((useconds_t)(release_interval_ms * 1000));
 		if (shutting_down)
 			continue; /* never touch HIP during teardown */
+		mem_mib = nvs_sum_allocated_mib();
+		if (mem_mib != last_mem_mib) {
+			struct nvs_msg m;
+			char buf[NVS_MSG_DATA_LEN];
+
+			snprintf(buf, sizeof(buf), "%ld", mem_mib);
+			nvs_msg_init(&m, NVS_MEM_UPDATE, client_id, buf);
+			pthread_mutex_lock(&sock_mutex);
+			if (sock_fd >= 0 &&
+			    nvs_send_msg(sock_fd, &m) == 0)
+				last_mem_mib = mem_mib;
+			pthread_mutex_unlock(&sock_mutex);
+		}
 		pthread_mutex_lock(&g_mutex);
 		have_lock = own_lock && scheduler_on;
 		worked = did_work;

@@ -93,4 +93,7 @@ void nvs_prefetch_allocs(void);
  * (NVSHARE_EVICT=1). */
 void nvs_evict_allocs(void);
 
+/* Sum of tracked allocations in MiB (hook.c). */
+long nvs_sum_allocated_mib(void);
+
 #endif /* NVSHARE_CLIENT_H */

@@ -61,17 +61,24 @@ static int send_one(uint8_t type, const char *data, int want_reply)
 		}
 		{
 			int on = 0, tq = 0, ncl = 0, qlen = 0;
+			long mib = 0;
 			char buf[NVS_MSG_DATA_LEN];
+			int nf;
 
 			memcpy(buf, m.data, NVS_MSG_DATA_LEN);
 			buf[NVS_MSG_DATA_LEN - 1] = '\0';
-			if (sscanf(buf, "%d,%d,%d,%d", &on, &tq, &ncl,
-				   &qlen) == 4)
+			nf = sscanf(buf, "%d,%d,%d,%d,%ld", &on, &tq,
+				    &ncl, &qlen, &mib);
+			if (nf >= 4) {
 				printf("scheduling: %s\ntq: %d s\n"
 				       "clients: %d\nqueued: %d\n",
 				       on ? "on" : "off", tq, ncl, qlen);
-			else
+				if (nf >= 5)
+					printf("tracked memory: %ld MiB\n",
+					       mib);
+			} else {
 				printf("status: %s\n", buf);
+			}
 		}
 	}
 	close(fd);

@@ -334,6 +334,16 @@ static void track_alloc(void *ptr, size_t size, int populated)
 		  populated ? "" : ", population deferred");
 }
 
+long nvs_sum_allocated_mib(void)
+{
+	size_t n;
+
+	pthread_mutex_lock(&alloc_mutex);
+	n = sum_allocated;
+	pthread_mutex_unlock(&alloc_mutex);
+	return (long)(n / NVS_MIB);
+}
+
 /* Called from the gate with the submission read lock held and the GPU
  * lock owned: materialize any deferred ranges before real work. */
 void nvs_populate_pending(void)

@@ -24,6 +24,7 @@ static const char *type_names[] = {
 	[NVS_SET_TQ] = "SET_TQ",
 	[NVS_STATUS_REQ] = "STATUS_REQ",
 	[NVS_STATUS] = "STATUS",
+	[NVS_MEM_UPDATE] = "MEM_UPDATE",
 };
 
 const char *nvs_msg_type_str(uint8_t type)

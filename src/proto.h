@@ -34,9 +34,10 @@ enum nvs_msg_type {
 	/* nvshare-amd extensions */
 	NVS_STATUS_REQ    = 9,
 	NVS_STATUS        = 10,
+	NVS_MEM_UPDATE    = 11,  /* client -> scheduler: tracked MiB */
 };
 
-#define NVS_MSG_TYPE_MAX NVS_STATUS
+#define NVS_MSG_TYPE_MAX NVS_MEM_UPDATE
 
 struct nvs_msg {
 	uint8_t type;

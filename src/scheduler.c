@@ -59,6 +59,7 @@ struct conn {
 	char pod_namespace[NVS_POD_NS_LEN];
 	int wants_lock;           /* present in its GPU's request queue */
 	int relock_pending;       /* holder re-requested during release */
+	long mem_mib;             /* client-reported tracked allocations */
 	struct conn *next;        /* registry list */
 	struct conn *qnext;       /* FCFS queue list */
 };
@@ -291,14 +292,26 @@ static void delete_conn(struct conn *c)
 	try_schedule(g);
 }
 
+static long total_mem_mib(void)
+{
+	long n = 0;
+	struct conn *c;
+
+	for (c = clients; c != NULL; c = c->next)
+		if (c->registered)
+			n += c->mem_mib;
+	return n;
+}
+
 static void handle_status_req(struct conn *c)
 {
 	struct nvs_msg m;
 	char buf[NVS_MSG_DATA_LEN];
 
-	/* data: "<on>,<tq>,<nclients>,<qlen>" */
-	snprintf(buf, sizeof(buf), "%d,%d,%d,%d", scheduler_on, tq_seconds,
-		 client_count(), queue_len_all());
+	/* data: "<on>,<tq>,<nclients>,<qlen>[,<mem_mib>]" */
+	snprintf(buf, sizeof(buf), "%d,%d,%d,%d,%ld", scheduler_on,
+		 tq_seconds, client_count(), queue_len_all(),
+		 total_mem_mib());
 	nvs_msg_init(&m, NVS_STATUS, c->id, buf);
 	if (nvs_send_msg(c->fd, &m) != 0)
 		log_warn("STATUS reply failed");
@@ -442,6 +455,17 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 			}
 		}
 		break;
+	case NVS_MEM_UPDATE: {
+		char buf[NVS_MSG_DATA_LEN];
+		long v;
+
+		memcpy(buf, m->data, NVS_MSG_DATA_LEN);
+		buf[NVS_MSG_DATA_LEN - 1] = '\0';
+		v = strtol(buf, NULL, 10);
+		if (c->registered && v >= 0)
+			c->mem_mib = v;
+		break;
+	}
 	case NVS_STATUS_REQ:
 		handle_status_req(c);
 		break;
@@ -499,8 +523,8 @@ static void dump_state(void)
 	}
 	for (c = clients; c != NULL; c = c->next)
 		log_info("  conn fd=%d reg=%d id=%016" PRIx64 " gpu=%d "
-			 "queued=%d%s", c->fd, c->registered, c->id,
-			 c->gpu, c->wants_lock,
+			 "queued=%d mem=%ldMiB%s", c->fd, c->registered,
+			 c->id, c->gpu, c->wants_lock, c->mem_mib,
 			 gpus[c->gpu].lock_holder == c ? " [HOLDER]" : "");
 }
 

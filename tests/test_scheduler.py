@@ -169,6 +169,23 @@ def test_status_counts(sched, sock_dir):
     assert st.clients == 0
 
 
+def test_mem_update_reported_in_status(sched, sock_dir):
+    a = make_client(sock_dir, "a")
+    b = make_client(sock_dir, "b")
+    a.send(proto.MEM_UPDATE, "2048")
+    b.send(proto.MEM_UPDATE, "1024")
+    time.sleep(0.2)
+    st = ctl.status(sock_dir)
+    assert st.tracked_mib == 3072
+    a.send(proto.MEM_UPDATE, "512")
+    time.sleep(0.2)
+    assert ctl.status(sock_dir).tracked_mib == 1536
+    a.close()
+    time.sleep(0.2)
+    assert ctl.status(sock_dir).tracked_mib == 1024
+    b.close()
+
+
 def test_stale_lock_released_ignored(sched, sock_dir):
     a = make_client(sock_dir, "a")
     b = make_client(sock_dir, "b")
