@@ -83,6 +83,6 @@ long nvs_env_long(const char *name, long dflt, long lo, long hi);
 /* Env var as boolean (unset/"0"/"" => 0, anything else => 1). */
 int nvs_env_bool(const char *name, int dflt);
 
-#define NVS_MIB (1024ULL * 1024ULL)
+#define NVS_MIB ((size_t)1024 * 1024)
 
 #endif /* NVSHARE_COMMON_H */
