@@ -306,12 +306,14 @@ static long total_mem_mib(void)
 static void handle_status_req(struct conn *c)
 {
 	struct nvs_msg m;
-	char buf[NVS_MSG_DATA_LEN];
+	char big[64], buf[NVS_MSG_DATA_LEN];
 
-	/* data: "<on>,<tq>,<nclients>,<qlen>[,<mem_mib>]" */
-	snprintf(buf, sizeof(buf), "%d,%d,%d,%d,%ld", scheduler_on,
+	/* data: "<on>,<tq>,<nclients>,<qlen>[,<mem_mib>]" — the wire
+	 * field is 20 bytes; mem_mib is last so only it can truncate. */
+	snprintf(big, sizeof(big), "%d,%d,%d,%d,%ld", scheduler_on,
 		 tq_seconds, client_count(), queue_len_all(),
 		 total_mem_mib());
+	nvs_strlcpy(buf, big, sizeof(buf));
 	nvs_msg_init(&m, NVS_STATUS, c->id, buf);
 	if (nvs_send_msg(c->fd, &m) != 0)
 		log_warn("STATUS reply failed");
