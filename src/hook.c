@@ -576,6 +576,11 @@ static void after_launch(void)
 	kern_since_sync = 0;
 	pthread_mutex_unlock(&win_mutex);
 
+	/* If the lock was just lost, the DROP_LOCK drain will sync; a
+	 * sync here would block on the NEXT holder's kernels and poison
+	 * the window adaptation with their runtime. */
+	if (!nvs_can_submit_now())
+		return;
 	CHECK_REAL(hipDeviceSynchronize);
 	t0 = nvs_now_ns();
 	real.hipDeviceSynchronize();
