@@ -262,8 +262,13 @@ static int connect_and_register(void)
 static void handle_lock_ok(void)
 {
 	/* Restore the working set BEFORE opening the gate: demand
-	 * faults racing the bulk migration degrade both. */
-	if (nvs_env_bool("NVSHARE_PREFETCH", 0))
+	 * faults racing the bulk migration degrade both.  Default
+	 * (NVSHARE_AUTO_MIGRATE=1): always restore on grant — when the
+	 * set is already resident the prefetch is a no-op costing
+	 * microseconds; when it was evicted this is 25x faster than
+	 * demand refaulting (profiles/restorebench.json). */
+	if (nvs_env_bool("NVSHARE_PREFETCH",
+			 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
 		nvs_prefetch_allocs();
 	pthread_mutex_lock(&g_mutex);
 	own_lock = 1;
@@ -283,7 +288,11 @@ static void handle_drop_lock(void)
 	if (!had)
 		return; /* already released voluntarily */
 	drain_gpu();
-	if (nvs_env_bool("NVSHARE_EVICT", 0))
+	/* Eviction self-gates on real memory pressure (it is skipped
+	 * when free HBM already fits the tracked set), so the automatic
+	 * default is safe for fitting workloads. */
+	if (nvs_env_bool("NVSHARE_EVICT",
+			 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
 		nvs_evict_allocs(); /* blocking; under pressure only */
 	send_msg_type(NVS_LOCK_RELEASED);
 	log_debug("client: lock released after DROP_LOCK");
