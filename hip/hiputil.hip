@@ -201,7 +201,8 @@ int nvs_hip_prefetch_chunked(const void *p, size_t bytes, int device,
 				base + off, len, device, streams[s]);
 			if (e != hipSuccess) {
 				for (i = 0; i < n; i++)
-					hipStreamDestroy(streams[i]);
+					(void)hipStreamDestroy(
+						streams[i]);
 				return (int)e;
 			}
 			off += len;
@@ -210,7 +211,8 @@ int nvs_hip_prefetch_chunked(const void *p, size_t bytes, int device,
 	}
 	for (i = 0; i < n; i++) {
 		hipError_t e = hipStreamSynchronize(streams[i]);
-		hipStreamDestroy(streams[i]);
+
+		(void)hipStreamDestroy(streams[i]);
 		if (e != hipSuccess)
 			return (int)e;
 	}
