@@ -714,17 +714,37 @@ nvshipError_t hipExtMallocWithFlags(void **ptr, size_t size,
 	return hipMalloc(ptr, size);
 }
 
+/*
+ * Stream-ordered allocations: PyTorch/MIOpen allocate a transient
+ * workspace this way EVERY STEP (measured: 60 hipMallocFromPoolAsync +
+ * 60 frees over 60 ResNet steps).  Converting those to managed costs
+ * a populate plus a blocking stream-sync on free per step — the
+ * measured ~1.27x ResNet overhead.  Small stream-ordered allocations
+ * are therefore passed through to the real allocator (transient,
+ * never a shareable working set); only large ones
+ * (> NVSHARE_PASSTHROUGH_MIB, default 256) are converted and tracked.
+ */
+static int pool_passthrough(size_t size)
+{
+	static long thresh_mib = -1;
+
+	if (thresh_mib < 0)
+		thresh_mib = nvs_env_long("NVSHARE_PASSTHROUGH_MIB", 256,
+					  0, 1024 * 1024);
+	return !disable_um && size <= (size_t)thresh_mib * NVS_MIB;
+}
+
 nvshipError_t hipMallocAsync(void **ptr, size_t size, nvship_stream_t s)
 {
-	/* Stream-ordered alloc becomes an immediate managed alloc: the
-	 * pointer is valid earlier than required, which is safe. */
-	(void)s;
 	BOOTSTRAP();
 	BUMP(H_hipMallocAsync);
-	if (disable_um) {
+	if (disable_um ||
+	    (pool_passthrough(size) && real.hipMallocAsync != NULL)) {
 		CHECK_REAL(hipMallocAsync);
 		return real.hipMallocAsync(ptr, size, s);
 	}
+	/* Large stream-ordered alloc becomes an immediate managed alloc:
+	 * the pointer is valid earlier than required (safe). */
 	return hipMalloc(ptr, size);
 }
 
@@ -732,8 +752,15 @@ nvshipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
 				     nvship_mempool_t pool,
 				     nvship_stream_t s)
 {
-	(void)pool;
-	return hipMallocAsync(ptr, size, s);
+	BOOTSTRAP();
+	BUMP(H_hipMallocFromPoolAsync);
+	if (disable_um ||
+	    (pool_passthrough(size) &&
+	     real.hipMallocFromPoolAsync != NULL)) {
+		CHECK_REAL(hipMallocFromPoolAsync);
+		return real.hipMallocFromPoolAsync(ptr, size, pool, s);
+	}
+	return hipMalloc(ptr, size);
 }
 
 nvshipError_t hipFree(void *ptr)
