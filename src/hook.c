@@ -639,9 +639,30 @@ static const char *hook_names[H_COUNT_] = {
 };
 
 static unsigned long hook_counts[H_COUNT_];
+static unsigned long hook_ns[H_COUNT_];     /* NVSHARE_PROFILE_HOOKS */
+static int profile_hooks = -1;
 
 #define BUMP(id) __atomic_fetch_add(&hook_counts[id], 1UL, \
 				    __ATOMIC_RELAXED)
+
+/* Wall-time accounting per hook (sum of wrapper time incl. the real
+ * call): `NVSHARE_PROFILE_HOOKS=1 NVSHARE_DEBUG=1` dumps per-hook
+ * total ms at exit — the tool for attributing interposer overhead
+ * (docs/roadmap.md #1). */
+static inline int64_t prof_begin(void)
+{
+	if (profile_hooks < 0)
+		profile_hooks = nvs_env_bool("NVSHARE_PROFILE_HOOKS", 0);
+	return profile_hooks ? nvs_now_ns() : 0;
+}
+
+static inline void prof_end(int id, int64_t t0)
+{
+	if (t0 != 0)
+		__atomic_fetch_add(&hook_ns[id],
+				   (unsigned long)(nvs_now_ns() - t0),
+				   __ATOMIC_RELAXED);
+}
 
 __attribute__((destructor)) static void dump_hook_counts(void)
 {
@@ -663,6 +684,23 @@ __attribute__((destructor)) static void dump_hook_counts(void)
 	}
 	if (any)
 		log_debug("hook call counts: %s", line);
+	if (profile_hooks == 1) {
+		off = 0;
+		line[0] = '\0';
+		for (i = 0; i < H_COUNT_; i++) {
+			if (hook_ns[i] == 0)
+				continue;
+			off += (size_t)snprintf(line + off,
+						sizeof(line) - off,
+						"%s%s=%.1fms",
+						off ? " " : "",
+						hook_names[i],
+						hook_ns[i] / 1e6);
+			if (off >= sizeof(line) - 64)
+				break;
+		}
+		log_debug("hook wall time: %s", line);
+	}
 }
 
 /* ------------------------------------------------------------------ */
@@ -823,24 +861,26 @@ nvshipError_t hipSetDevice(int dev)
 
 /* ---- gated work submissions ---- */
 
-#define GATED(call)                                                        \
+#define GATED2(id, call)                                                   \
 	do {                                                               \
 		nvshipError_t r_;                                          \
-		BOOTSTRAP();                                               \
+		int64_t t0_ = prof_begin();                                \
 		nvs_submit_begin();                                        \
 		r_ = (call);                                               \
 		nvs_submit_end();                                          \
+		prof_end(id, t0_);                                         \
 		return r_;                                                 \
 	} while (0)
 
-#define GATED_LAUNCH(call)                                                 \
+#define GATED_LAUNCH2(id, call)                                            \
 	do {                                                               \
 		nvshipError_t r_;                                          \
-		BOOTSTRAP();                                               \
+		int64_t t0_ = prof_begin();                                \
 		nvs_submit_begin();                                        \
 		r_ = (call);                                               \
 		nvs_submit_end();                                          \
 		after_launch();                                            \
+		prof_end(id, t0_);                                         \
 		return r_;                                                 \
 	} while (0)
 
@@ -851,7 +891,7 @@ nvshipError_t hipLaunchKernel(const void *f, nvship_dim3 grid,
 	BOOTSTRAP();
 	BUMP(H_hipLaunchKernel);
 	CHECK_REAL(hipLaunchKernel);
-	GATED_LAUNCH(real.hipLaunchKernel(f, grid, block, args, shmem,
+	GATED_LAUNCH2(H_hipLaunchKernel, real.hipLaunchKernel(f, grid, block, args, shmem,
 					  stream));
 }
 
@@ -864,7 +904,7 @@ nvshipError_t hipExtLaunchKernel(const void *f, nvship_dim3 grid,
 	BOOTSTRAP();
 	BUMP(H_hipExtLaunchKernel);
 	CHECK_REAL(hipExtLaunchKernel);
-	GATED_LAUNCH(real.hipExtLaunchKernel(f, grid, block, args, shmem,
+	GATED_LAUNCH2(H_hipExtLaunchKernel, real.hipExtLaunchKernel(f, grid, block, args, shmem,
 					     stream, ev0, ev1, flags));
 }
 
@@ -876,7 +916,7 @@ nvshipError_t hipLaunchCooperativeKernel(const void *f, nvship_dim3 grid,
 	BOOTSTRAP();
 	BUMP(H_hipLaunchCooperativeKernel);
 	CHECK_REAL(hipLaunchCooperativeKernel);
-	GATED_LAUNCH(real.hipLaunchCooperativeKernel(f, grid, block, args,
+	GATED_LAUNCH2(H_hipLaunchCooperativeKernel, real.hipLaunchCooperativeKernel(f, grid, block, args,
 						     shmem, stream));
 }
 
@@ -890,7 +930,7 @@ nvshipError_t hipModuleLaunchKernel(nvship_function_t f, unsigned int gx,
 	BOOTSTRAP();
 	BUMP(H_hipModuleLaunchKernel);
 	CHECK_REAL(hipModuleLaunchKernel);
-	GATED_LAUNCH(real.hipModuleLaunchKernel(f, gx, gy, gz, bx, by, bz,
+	GATED_LAUNCH2(H_hipModuleLaunchKernel, real.hipModuleLaunchKernel(f, gx, gy, gz, bx, by, bz,
 						shmem, stream, params,
 						extra));
 }
@@ -907,7 +947,7 @@ nvshipError_t hipExtModuleLaunchKernel(nvship_function_t f, uint32_t gwx,
 	BOOTSTRAP();
 	BUMP(H_hipExtModuleLaunchKernel);
 	CHECK_REAL(hipExtModuleLaunchKernel);
-	GATED_LAUNCH(real.hipExtModuleLaunchKernel(f, gwx, gwy, gwz, lwx,
+	GATED_LAUNCH2(H_hipExtModuleLaunchKernel, real.hipExtModuleLaunchKernel(f, gwx, gwy, gwz, lwx,
 						   lwy, lwz, shmem, stream,
 						   params, extra, ev0, ev1,
 						   flags));
@@ -918,7 +958,7 @@ nvshipError_t hipGraphLaunch(nvship_graphexec_t g, nvship_stream_t stream)
 	BOOTSTRAP();
 	BUMP(H_hipGraphLaunch);
 	CHECK_REAL(hipGraphLaunch);
-	GATED_LAUNCH(real.hipGraphLaunch(g, stream));
+	GATED_LAUNCH2(H_hipGraphLaunch, real.hipGraphLaunch(g, stream));
 }
 
 nvshipError_t hipMemcpy(void *dst, const void *src, size_t n,
@@ -927,7 +967,7 @@ nvshipError_t hipMemcpy(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpy);
 	CHECK_REAL(hipMemcpy);
-	GATED(real.hipMemcpy(dst, src, n, kind));
+	GATED2(H_hipMemcpy, real.hipMemcpy(dst, src, n, kind));
 }
 
 nvshipError_t hipMemcpyAsync(void *dst, const void *src, size_t n,
@@ -936,7 +976,7 @@ nvshipError_t hipMemcpyAsync(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyAsync);
 	CHECK_REAL(hipMemcpyAsync);
-	GATED(real.hipMemcpyAsync(dst, src, n, kind, s));
+	GATED2(H_hipMemcpyAsync, real.hipMemcpyAsync(dst, src, n, kind, s));
 }
 
 nvshipError_t hipMemcpyWithStream(void *dst, const void *src, size_t n,
@@ -946,7 +986,7 @@ nvshipError_t hipMemcpyWithStream(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyWithStream);
 	CHECK_REAL(hipMemcpyWithStream);
-	GATED(real.hipMemcpyWithStream(dst, src, n, kind, s));
+	GATED2(H_hipMemcpyWithStream, real.hipMemcpyWithStream(dst, src, n, kind, s));
 }
 
 nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
@@ -955,7 +995,7 @@ nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyHtoD);
 	CHECK_REAL(hipMemcpyHtoD);
-	GATED(real.hipMemcpyHtoD(dst, src, n));
+	GATED2(H_hipMemcpyHtoD, real.hipMemcpyHtoD(dst, src, n));
 }
 
 nvshipError_t hipMemcpyDtoH(void *dst, nvship_deviceptr_t src, size_t n)
@@ -963,7 +1003,7 @@ nvshipError_t hipMemcpyDtoH(void *dst, nvship_deviceptr_t src, size_t n)
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyDtoH);
 	CHECK_REAL(hipMemcpyDtoH);
-	GATED(real.hipMemcpyDtoH(dst, src, n));
+	GATED2(H_hipMemcpyDtoH, real.hipMemcpyDtoH(dst, src, n));
 }
 
 nvshipError_t hipMemcpyDtoD(nvship_deviceptr_t dst, nvship_deviceptr_t src,
@@ -972,7 +1012,7 @@ nvshipError_t hipMemcpyDtoD(nvship_deviceptr_t dst, nvship_deviceptr_t src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyDtoD);
 	CHECK_REAL(hipMemcpyDtoD);
-	GATED(real.hipMemcpyDtoD(dst, src, n));
+	GATED2(H_hipMemcpyDtoD, real.hipMemcpyDtoD(dst, src, n));
 }
 
 nvshipError_t hipMemcpyHtoDAsync(nvship_deviceptr_t dst, const void *src,
@@ -981,7 +1021,7 @@ nvshipError_t hipMemcpyHtoDAsync(nvship_deviceptr_t dst, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyHtoDAsync);
 	CHECK_REAL(hipMemcpyHtoDAsync);
-	GATED(real.hipMemcpyHtoDAsync(dst, src, n, s));
+	GATED2(H_hipMemcpyHtoDAsync, real.hipMemcpyHtoDAsync(dst, src, n, s));
 }
 
 nvshipError_t hipMemcpyDtoHAsync(void *dst, nvship_deviceptr_t src,
@@ -990,7 +1030,7 @@ nvshipError_t hipMemcpyDtoHAsync(void *dst, nvship_deviceptr_t src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyDtoHAsync);
 	CHECK_REAL(hipMemcpyDtoHAsync);
-	GATED(real.hipMemcpyDtoHAsync(dst, src, n, s));
+	GATED2(H_hipMemcpyDtoHAsync, real.hipMemcpyDtoHAsync(dst, src, n, s));
 }
 
 nvshipError_t hipMemcpyDtoDAsync(nvship_deviceptr_t dst,
@@ -1000,7 +1040,7 @@ nvshipError_t hipMemcpyDtoDAsync(nvship_deviceptr_t dst,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyDtoDAsync);
 	CHECK_REAL(hipMemcpyDtoDAsync);
-	GATED(real.hipMemcpyDtoDAsync(dst, src, n, s));
+	GATED2(H_hipMemcpyDtoDAsync, real.hipMemcpyDtoDAsync(dst, src, n, s));
 }
 
 nvshipError_t hipMemset(void *dst, int value, size_t n)
@@ -1008,7 +1048,7 @@ nvshipError_t hipMemset(void *dst, int value, size_t n)
 	BOOTSTRAP();
 	BUMP(H_hipMemset);
 	CHECK_REAL(hipMemset);
-	GATED(real.hipMemset(dst, value, n));
+	GATED2(H_hipMemset, real.hipMemset(dst, value, n));
 }
 
 nvshipError_t hipMemsetAsync(void *dst, int value, size_t n,
@@ -1017,7 +1057,7 @@ nvshipError_t hipMemsetAsync(void *dst, int value, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemsetAsync);
 	CHECK_REAL(hipMemsetAsync);
-	GATED(real.hipMemsetAsync(dst, value, n, s));
+	GATED2(H_hipMemsetAsync, real.hipMemsetAsync(dst, value, n, s));
 }
 
 nvshipError_t hipMemsetD32Async(nvship_deviceptr_t dst, int value,
@@ -1026,7 +1066,7 @@ nvshipError_t hipMemsetD32Async(nvship_deviceptr_t dst, int value,
 	BOOTSTRAP();
 	BUMP(H_hipMemsetD32Async);
 	CHECK_REAL(hipMemsetD32Async);
-	GATED(real.hipMemsetD32Async(dst, value, count, s));
+	GATED2(H_hipMemsetD32Async, real.hipMemsetD32Async(dst, value, count, s));
 }
 
 nvshipError_t hipMemcpy2D(void *dst, size_t dpitch, const void *src,
@@ -1036,7 +1076,7 @@ nvshipError_t hipMemcpy2D(void *dst, size_t dpitch, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpy2D);
 	CHECK_REAL(hipMemcpy2D);
-	GATED(real.hipMemcpy2D(dst, dpitch, src, spitch, width, height,
+	GATED2(H_hipMemcpy2D, real.hipMemcpy2D(dst, dpitch, src, spitch, width, height,
 			       kind));
 }
 
@@ -1047,7 +1087,7 @@ nvshipError_t hipMemcpy2DAsync(void *dst, size_t dpitch, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpy2DAsync);
 	CHECK_REAL(hipMemcpy2DAsync);
-	GATED(real.hipMemcpy2DAsync(dst, dpitch, src, spitch, width,
+	GATED2(H_hipMemcpy2DAsync, real.hipMemcpy2DAsync(dst, dpitch, src, spitch, width,
 				    height, kind, s));
 }
 
@@ -1058,7 +1098,7 @@ nvshipError_t hipMemcpyToSymbol(const void *symbol, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyToSymbol);
 	CHECK_REAL(hipMemcpyToSymbol);
-	GATED(real.hipMemcpyToSymbol(symbol, src, n, offset, kind));
+	GATED2(H_hipMemcpyToSymbol, real.hipMemcpyToSymbol(symbol, src, n, offset, kind));
 }
 
 nvshipError_t hipMemcpyFromSymbol(void *dst, const void *symbol, size_t n,
@@ -1067,7 +1107,7 @@ nvshipError_t hipMemcpyFromSymbol(void *dst, const void *symbol, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyFromSymbol);
 	CHECK_REAL(hipMemcpyFromSymbol);
-	GATED(real.hipMemcpyFromSymbol(dst, symbol, n, offset, kind));
+	GATED2(H_hipMemcpyFromSymbol, real.hipMemcpyFromSymbol(dst, symbol, n, offset, kind));
 }
 
 nvshipError_t hipMemcpyPeerAsync(void *dst, int dst_dev, const void *src,
@@ -1076,7 +1116,7 @@ nvshipError_t hipMemcpyPeerAsync(void *dst, int dst_dev, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyPeerAsync);
 	CHECK_REAL(hipMemcpyPeerAsync);
-	GATED(real.hipMemcpyPeerAsync(dst, dst_dev, src, src_dev, n, s));
+	GATED2(H_hipMemcpyPeerAsync, real.hipMemcpyPeerAsync(dst, dst_dev, src, src_dev, n, s));
 }
 
 nvshipError_t hipModuleLaunchCooperativeKernel(
@@ -1087,7 +1127,7 @@ nvshipError_t hipModuleLaunchCooperativeKernel(
 	BOOTSTRAP();
 	BUMP(H_hipModuleLaunchCooperativeKernel);
 	CHECK_REAL(hipModuleLaunchCooperativeKernel);
-	GATED_LAUNCH(real.hipModuleLaunchCooperativeKernel(
+	GATED_LAUNCH2(H_hipModuleLaunchCooperativeKernel, real.hipModuleLaunchCooperativeKernel(
 		f, gx, gy, gz, bx, by, bz, shmem, stream, params));
 }
 
