@@ -169,6 +169,32 @@ def test_status_counts(sched, sock_dir):
     assert st.clients == 0
 
 
+def test_set_tq_extends_current_quantum(sched, sock_dir):
+    """Raising TQ mid-quantum defers the pending preemption."""
+    a = make_client(sock_dir, "a")
+    b = make_client(sock_dir, "b")
+    a.send(proto.REQ_LOCK)
+    assert a.recv(5).type == proto.LOCK_OK
+    b.send(proto.REQ_LOCK)          # arms the TQ=1 preemption
+    ctl.set_tq(30, sock_dir)        # extend before it fires
+    with pytest.raises(Exception):
+        a.recv(2.5)                 # no DROP_LOCK under the new TQ
+    ctl.set_tq(1, sock_dir)         # shrink back: holder is overdue
+    assert a.recv(5).type == proto.DROP_LOCK
+    a.send(proto.LOCK_RELEASED)
+    assert b.recv(5).type == proto.LOCK_OK
+    a.close()
+    b.close()
+
+
+def test_mem_update_from_unregistered_ignored(sched, sock_dir):
+    s = proto.Client(sock_dir=sock_dir, pod_name="raw").connect()
+    proto.send_msg(s.sock, proto.Message(proto.MEM_UPDATE, data="999"))
+    time.sleep(0.2)
+    assert ctl.status(sock_dir).tracked_mib == 0
+    s.close()
+
+
 def test_mem_update_reported_in_status(sched, sock_dir):
     a = make_client(sock_dir, "a")
     b = make_client(sock_dir, "b")
