@@ -142,6 +142,55 @@ def test_register_with_fake_kubelet(plugin, kubelet_dir):
         server.stop(grace=1)
 
 
+def test_restart_on_kubelet_socket_recreation(kubelet_dir):
+    """run_forever re-registers when the kubelet socket is recreated
+    (reference main.go:132-177 inotify restart loop)."""
+    registrations = []
+    ev = threading.Event()
+
+    def register(req_bytes, ctx):
+        registrations.append(pb.RegisterRequest.FromString(req_bytes))
+        ev.set()
+        return pb.Empty()
+
+    handler = grpc.method_handlers_generic_handler(
+        pb.REGISTRATION_SERVICE,
+        {"Register": grpc.unary_unary_rpc_method_handler(
+            register,
+            response_serializer=lambda m: m.SerializeToString())})
+
+    def make_kubelet():
+        server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+        server.add_generic_rpc_handlers((handler,))
+        server.add_insecure_port(
+            f"unix://{kubelet_dir}/kubelet.sock")
+        server.start()
+        return server
+
+    k1 = make_kubelet()
+    cfg = PluginConfig(virtual_devices=2, kubelet_dir=kubelet_dir,
+                       gpus=[0])
+    p = DevicePlugin(cfg)
+    t = threading.Thread(target=p.run_forever, daemon=True)
+    t.start()
+    try:
+        assert ev.wait(10), "first registration missing"
+        ev.clear()
+        n_first = len(registrations)
+        # Recreate the kubelet socket (inode changes).
+        k1.stop(grace=0)
+        import contextlib
+        import os as _os
+        with contextlib.suppress(FileNotFoundError):
+            _os.unlink(f"{kubelet_dir}/kubelet.sock")
+        k2 = make_kubelet()
+        assert ev.wait(15), "no re-registration after recreation"
+        assert len(registrations) > n_first
+        k2.stop(grace=0)
+    finally:
+        p.stop()
+
+
 def test_preferred_allocation(plugin):
     req = pb.PreferredAllocationRequest()
     creq = req.container_requests.add()
