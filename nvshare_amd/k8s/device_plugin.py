@@ -210,6 +210,7 @@ class DevicePlugin:
         self.cfg = cfg or PluginConfig()
         self.servicer = DevicePluginServicer(self.cfg)
         self.server: grpc.Server | None = None
+        self._shutdown = threading.Event()
 
     @property
     def endpoint(self) -> str:
@@ -244,6 +245,7 @@ class DevicePlugin:
         log.info("registered %s with kubelet", self.cfg.resource_name)
 
     def stop(self) -> None:
+        self._shutdown.set()
         self.servicer.stop()
         if self.server is not None:
             self.server.stop(grace=1)
@@ -252,21 +254,25 @@ class DevicePlugin:
     def run_forever(self) -> None:
         """Serve + register; restart on kubelet socket recreation."""
         kubelet = os.path.join(self.cfg.kubelet_dir, KUBELET_SOCK)
-        while True:
+        while not self._shutdown.is_set():
             self.start()
             try:
                 self.register()
             except grpc.RpcError as e:
                 log.error("kubelet registration failed: %s", e)
-                self.stop()
-                time.sleep(5)
+                self.servicer.stop()
+                if self.server is not None:
+                    self.server.stop(grace=1)
+                    self.server = None
+                self._shutdown.wait(5)
                 continue
             try:
                 ino = os.stat(kubelet).st_ino
             except FileNotFoundError:
                 ino = None
-            while True:
-                time.sleep(2)
+            while not self._shutdown.is_set():
+                if self._shutdown.wait(2):
+                    break
                 try:
                     now = os.stat(kubelet).st_ino
                 except FileNotFoundError:
@@ -274,7 +280,13 @@ class DevicePlugin:
                 if now != ino:
                     log.warning("kubelet socket changed; restarting")
                     break
-            self.stop()
+            if not self._shutdown.is_set():
+                self.servicer.stop()
+                if self.server is not None:
+                    self.server.stop(grace=1)
+                    self.server = None
+                # a fresh servicer for the next serve cycle
+                self.servicer = DevicePluginServicer(self.cfg)
 
 
 def main() -> None:
