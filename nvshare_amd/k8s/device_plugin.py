@@ -257,27 +257,29 @@ class DevicePlugin:
         while not self._shutdown.is_set():
             self.start()
             try:
-                self.register()
-            except grpc.RpcError as e:
+                self.register(timeout=5)
+            except Exception as e:  # noqa: BLE001 — keep serving loop alive
                 log.error("kubelet registration failed: %s", e)
                 self.servicer.stop()
                 if self.server is not None:
                     self.server.stop(grace=1)
                     self.server = None
-                self._shutdown.wait(5)
+                self._shutdown.wait(2)
                 continue
-            try:
-                ino = os.stat(kubelet).st_ino
-            except FileNotFoundError:
-                ino = None
+            def socket_id():
+                # inode numbers get reused quickly on tmpfs; ctime
+                # disambiguates a same-inode recreation.
+                try:
+                    st = os.stat(kubelet)
+                    return (st.st_ino, st.st_ctime_ns)
+                except FileNotFoundError:
+                    return None
+
+            ino = socket_id()
             while not self._shutdown.is_set():
                 if self._shutdown.wait(2):
                     break
-                try:
-                    now = os.stat(kubelet).st_ino
-                except FileNotFoundError:
-                    now = None
-                if now != ino:
+                if socket_id() != ino:
                     log.warning("kubelet socket changed; restarting")
                     break
             if not self._shutdown.is_set():

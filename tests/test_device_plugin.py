@@ -10,6 +10,7 @@ from __future__ import annotations
 
 import tempfile
 import threading
+import time
 from concurrent import futures
 
 import grpc
@@ -177,6 +178,9 @@ def test_restart_on_kubelet_socket_recreation(kubelet_dir):
         assert ev.wait(10), "first registration missing"
         ev.clear()
         n_first = len(registrations)
+        # Let the plugin snapshot the current socket inode before we
+        # swap it (it stats right after register returns).
+        time.sleep(1.0)
         # Recreate the kubelet socket (inode changes).
         k1.stop(grace=0)
         import contextlib
