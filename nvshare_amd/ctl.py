@@ -37,10 +37,12 @@ def _one_shot(msg_type: int, data: str = "",
         s.close()
 
 
-def set_tq(seconds: int, sock_dir: str | None = None) -> None:
+def set_tq(seconds: int, sock_dir: str | None = None,
+           gpu: int | None = None) -> None:
     if not 1 <= seconds <= 86400:
         raise ValueError("TQ must be in [1, 86400] seconds")
-    _one_shot(proto.SET_TQ, str(seconds), sock_dir)
+    data = f"gpu{gpu}:{seconds}" if gpu is not None else str(seconds)
+    _one_shot(proto.SET_TQ, data, sock_dir)
 
 
 def set_scheduling(on: bool, sock_dir: str | None = None) -> None:

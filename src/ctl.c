@@ -22,6 +22,8 @@ static void usage(const char *argv0)
 		"Usage: %s [OPTION]...\n"
 		"Configure the nvshare-amd scheduler.\n\n"
 		"  -T, --set-tq SECONDS    set the time quantum (>= 1)\n"
+		"  -g, --gpu N             apply -T to one GPU only "
+		"(put -g before -T)\n"
 		"  -S, --scheduler on|off  enable/disable anti-thrashing "
 		"scheduling\n"
 		"  -q, --status            print scheduler status\n"
@@ -90,19 +92,25 @@ int main(int argc, char **argv)
 	static const struct option longopts[] = {
 		{ "set-tq", required_argument, NULL, 'T' },
 		{ "scheduler", required_argument, NULL, 'S' },
+		{ "gpu", required_argument, NULL, 'g' },
 		{ "status", no_argument, NULL, 'q' },
 		{ "help", no_argument, NULL, 'h' },
 		{ NULL, 0, NULL, 0 },
 	};
 	int opt, did_something = 0, rc = 0;
+	long gpu = -1;
 
 	nvs_log_init();
-	while ((opt = getopt_long(argc, argv, "T:S:qh", longopts, NULL))
+	while ((opt = getopt_long(argc, argv, "T:S:g:qh", longopts, NULL))
 	       != -1) {
 		switch (opt) {
+		case 'g':
+			gpu = strtol(optarg, NULL, 10);
+			break;
 		case 'T': {
 			char *end = NULL;
 			long v = strtol(optarg, &end, 10);
+			char data[32];
 
 			if (end == optarg || *end != '\0' || v < 1 ||
 			    v > 86400) {
@@ -111,7 +119,12 @@ int main(int argc, char **argv)
 					optarg);
 				return 1;
 			}
-			rc |= send_one(NVS_SET_TQ, optarg, 0);
+			if (gpu >= 0)
+				snprintf(data, sizeof(data), "gpu%ld:%ld",
+					 gpu, v);
+			else
+				snprintf(data, sizeof(data), "%ld", v);
+			rc |= send_one(NVS_SET_TQ, data, 0);
 			did_something = 1;
 			break;
 		}

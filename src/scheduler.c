@@ -74,6 +74,7 @@ struct gpu_state {
 	unsigned long round;
 	int64_t quantum_start_ns;
 	unsigned long grants, preemptions;
+	int tq_override;          /* 0 = use the global tq */
 };
 
 static struct conn *clients;      /* all connections */
@@ -182,6 +183,11 @@ static void try_schedule(struct gpu_state *g)
 
 /* A GPU is "armed" when its holder can be preempted: lock held, at
  * least one waiter behind the holder, DROP_LOCK not yet sent. */
+static int gpu_tq(const struct gpu_state *g)
+{
+	return g->tq_override > 0 ? g->tq_override : tq_seconds;
+}
+
 static int gpu_armed(const struct gpu_state *g)
 {
 	return g->lock_held && scheduler_on && !g->drop_lock_sent &&
@@ -213,7 +219,7 @@ static void *timer_thread(void *arg)
 				continue;
 			any_armed = 1;
 			deadline = g->quantum_start_ns +
-				   (int64_t)tq_seconds * 1000000000LL;
+				   (int64_t)gpu_tq(g) * 1000000000LL;
 			if (deadline <= now) {
 				g->drop_lock_sent = 1;
 				g->preemptions++;
@@ -409,16 +415,38 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 		char buf[NVS_MSG_DATA_LEN];
 		long v;
 		char *end = NULL;
+		int gpu = -1;
+		const char *num = buf;
 
 		memcpy(buf, m->data, NVS_MSG_DATA_LEN);
 		buf[NVS_MSG_DATA_LEN - 1] = '\0';
-		v = strtol(buf, &end, 10);
-		if (end == buf || v < 1 || v > 86400) {
+		/* "N" sets the global TQ; "gpuK:N" overrides one GPU. */
+		if (strncmp(buf, "gpu", 3) == 0) {
+			char *colon = strchr(buf, ':');
+
+			if (colon == NULL) {
+				log_warn("SET_TQ: malformed '%s'", buf);
+				break;
+			}
+			gpu = (int)strtol(buf + 3, NULL, 10);
+			if (gpu < 0 || gpu >= MAX_GPUS) {
+				log_warn("SET_TQ: bad gpu in '%s'", buf);
+				break;
+			}
+			num = colon + 1;
+		}
+		v = strtol(num, &end, 10);
+		if (end == num || v < 1 || v > 86400) {
 			log_warn("SET_TQ: invalid value '%s'", buf);
 			break;
 		}
-		tq_seconds = (int)v;
-		log_info("TQ set to %d s", tq_seconds);
+		if (gpu >= 0) {
+			gpus[gpu].tq_override = (int)v;
+			log_info("TQ for gpu%d set to %d s", gpu, (int)v);
+		} else {
+			tq_seconds = (int)v;
+			log_info("TQ set to %d s", tq_seconds);
+		}
 		pthread_cond_broadcast(&timer_cv);
 		break;
 	}

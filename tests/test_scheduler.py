@@ -276,3 +276,25 @@ def test_env_sched_off_startup(artifacts, sock_dir):
         reply = c.register()
         assert reply.type == proto.SCHED_OFF
         c.close()
+
+
+def test_per_gpu_tq_override(sched, sock_dir):
+    """SET_TQ "gpuN:tq" overrides one GPU's quantum only."""
+    ctl.set_tq(30, sock_dir)           # global: long
+    ctl.set_tq(1, sock_dir, gpu=1)     # gpu1: short
+    a0 = make_client(sock_dir, "a0", gpu=0)
+    b0 = make_client(sock_dir, "b0", gpu=0)
+    a1 = make_client(sock_dir, "a1", gpu=1)
+    b1 = make_client(sock_dir, "b1", gpu=1)
+    a0.send(proto.REQ_LOCK)
+    assert a0.recv(5).type == proto.LOCK_OK
+    a1.send(proto.REQ_LOCK)
+    assert a1.recv(5).type == proto.LOCK_OK
+    b0.send(proto.REQ_LOCK)
+    b1.send(proto.REQ_LOCK)
+    # gpu1's holder preempts after ~1 s; gpu0's not within 3 s.
+    assert a1.recv(5).type == proto.DROP_LOCK
+    with pytest.raises(Exception):
+        a0.recv(2.0)
+    for c in (a0, b0, a1, b1):
+        c.close()
