@@ -161,6 +161,55 @@ def test_two_clients_serialized(artifacts, sched, sock_dir):
     assert "DROP_LOCK" in sched_log
 
 
+def test_pool_alloc_passthrough_threshold(artifacts, sched, sock_dir):
+    """Small stream-ordered allocs pass through to the real allocator;
+    large ones convert to managed (NVSHARE_PASSTHROUGH_MIB)."""
+    import ctypes
+
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     reserve_mib=64, oversubscribe=True)
+    env["NVSTUB_TOTAL_MIB"] = "4096"
+    log = os.path.join(sock_dir, "pool.log")
+    env["NVSTUB_LOG"] = log
+    env["NVSHARE_PASSTHROUGH_MIB"] = "8"
+    # Drive hipMallocAsync directly through a preloaded helper.
+    code = r"""
+#include <stdio.h>
+#include <stdlib.h>
+typedef int hipError_t;
+extern hipError_t hipMallocAsync(void **, unsigned long, void *);
+extern hipError_t hipFreeAsync(void *, void *);
+extern hipError_t hipDeviceSynchronize(void);
+int main(void) {
+    void *small = 0, *big = 0;
+    if (hipMallocAsync(&small, 4ul << 20, 0) != 0) return 1;
+    if (hipMallocAsync(&big, 64ul << 20, 0) != 0) return 2;
+    hipFreeAsync(small, 0);
+    hipFreeAsync(big, 0);
+    hipDeviceSynchronize();
+    puts("POOL_OK");
+    return 0;
+}
+"""
+    src = os.path.join(sock_dir, "pool.c")
+    exe = os.path.join(sock_dir, "pool")
+    with open(src, "w") as f:
+        f.write(code)
+    build = subprocess.run(
+        ["gcc", "-o", exe, src, "-L", str(artifacts.stub_dir),
+         "-lamdhip64", f"-Wl,-rpath,{artifacts.stub_dir}"],
+        capture_output=True, text=True)
+    assert build.returncode == 0, build.stderr
+    r = subprocess.run([exe], env=env, capture_output=True, text=True,
+                       timeout=60)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    names = [e[2] for e in load_events(log)]
+    # 4 MiB <= 8 MiB threshold: forwarded as a real async alloc.
+    assert "hipMallocAsync" in names
+    # 64 MiB > threshold: converted to managed.
+    assert "hipMallocManaged" in names
+
+
 def test_fork_safety(artifacts, sched, sock_dir):
     """A forked child free-runs without corrupting the parent's
     scheduler protocol or deadlocking (reference would deadlock)."""
