@@ -12,7 +12,7 @@ import json
 import subprocess
 import sys
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 from nvshare_amd.env import client_env
 

@@ -10,7 +10,6 @@ Validates the MI355X-specific risks SURVEY.md §7 ranks:
 from __future__ import annotations
 
 import ctypes
-import json
 import os
 import subprocess
 import sys

@@ -12,9 +12,6 @@ from __future__ import annotations
 import os
 import subprocess
 import time
-from collections import defaultdict
-
-import pytest
 
 from nvshare_amd.env import client_env
 
@@ -164,8 +161,6 @@ def test_two_clients_serialized(artifacts, sched, sock_dir):
 def test_pool_alloc_passthrough_threshold(artifacts, sched, sock_dir):
     """Small stream-ordered allocs pass through to the real allocator;
     large ones convert to managed (NVSHARE_PASSTHROUGH_MIB)."""
-    import ctypes
-
     env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
                      reserve_mib=64, oversubscribe=True)
     env["NVSTUB_TOTAL_MIB"] = "4096"
