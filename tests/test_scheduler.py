@@ -72,7 +72,7 @@ def test_tq_preemption(sched, sock_dir):
     msg = a.recv(5)
     dt = time.monotonic() - t0
     assert msg.type == proto.DROP_LOCK
-    assert 0.3 < dt < 3.0, f"preemption after {dt:.2f}s with TQ=1"
+    assert 0.3 < dt < 4.5, f"preemption after {dt:.2f}s with TQ=1"
     a.send(proto.LOCK_RELEASED)
     assert b.recv(5).type == proto.LOCK_OK
     a.close()
@@ -101,7 +101,7 @@ def test_waiter_arrival_preempts_overdue_holder(sched, sock_dir):
     msg = a.recv(5)
     dt = time.monotonic() - t0
     assert msg.type == proto.DROP_LOCK
-    assert dt < 1.0, f"overdue holder preempted after {dt:.2f}s"
+    assert dt < 2.5, f"overdue holder preempted after {dt:.2f}s"
     a.close()
     b.close()
 
