@@ -42,3 +42,26 @@ def test_truncation():
 def test_type_names():
     assert proto.TYPE_NAMES[1] == "REGISTER"
     assert proto.TYPE_NAMES[8] == "SET_TQ"
+
+
+def test_reference_shaped_client_compat(sched, sock_dir):
+    """A reference-shaped client (REGISTER with empty data, no
+    extension types) gets the full lock lifecycle on gpu0."""
+    import socket as socklib
+
+    from nvshare_amd import proto
+
+    s = socklib.socket(socklib.AF_UNIX, socklib.SOCK_STREAM)
+    s.settimeout(5)
+    s.connect(proto.scheduler_path(sock_dir))
+    # exactly what the reference sends: type=REGISTER, empty data
+    proto.send_msg(s, proto.Message(proto.REGISTER, "refpod", "refns"))
+    reply = proto.recv_msg(s, 5)
+    assert reply.type == proto.SCHED_ON
+    assert len(reply.data) == 16  # hex client id
+    proto.send_msg(s, proto.Message(proto.REQ_LOCK))
+    assert proto.recv_msg(s, 5).type == proto.LOCK_OK
+    proto.send_msg(s, proto.Message(proto.LOCK_RELEASED))
+    proto.send_msg(s, proto.Message(proto.REQ_LOCK))
+    assert proto.recv_msg(s, 5).type == proto.LOCK_OK
+    s.close()
