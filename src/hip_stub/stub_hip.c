@@ -333,6 +333,68 @@ hipError_t hipMemcpyDtoDAsync(void *d, void *s, size_t n, hipStream_t st)
 	return do_copy(d, s, n);
 }
 
+hipError_t hipMemcpy2D(void *d, size_t dp, const void *s, size_t sp,
+		       size_t w, size_t h, int k)
+{
+	(void)dp; (void)sp; (void)k;
+	return do_copy(d, s, w * h);
+}
+
+hipError_t hipMemcpy2DAsync(void *d, size_t dp, const void *s, size_t sp,
+			    size_t w, size_t h, int k, hipStream_t st)
+{
+	(void)dp; (void)sp; (void)k; (void)st;
+	return do_copy(d, s, w * h);
+}
+
+hipError_t hipMemcpyToSymbol(const void *sym, const void *s, size_t n,
+			     size_t off, int k)
+{
+	(void)sym; (void)s; (void)n; (void)off; (void)k;
+	ev("memcpyToSymbol", (long long)n);
+	return 0;
+}
+
+hipError_t hipMemcpyFromSymbol(void *d, const void *sym, size_t n,
+			       size_t off, int k)
+{
+	(void)d; (void)sym; (void)n; (void)off; (void)k;
+	ev("memcpyFromSymbol", (long long)n);
+	return 0;
+}
+
+hipError_t hipMemcpyPeerAsync(void *d, int dd, const void *s, int sd,
+			      size_t n, hipStream_t st)
+{
+	(void)dd; (void)sd; (void)st;
+	return do_copy(d, s, n);
+}
+
+hipError_t hipMemsetD32Async(void *d, int v, size_t count, hipStream_t s)
+{
+	(void)s;
+	ev("memsetD32", (long long)count);
+	if (d != NULL) {
+		int *p = d;
+		size_t i;
+
+		for (i = 0; i < count; i++)
+			p[i] = v;
+	}
+	return 0;
+}
+
+hipError_t hipModuleLaunchCooperativeKernel(void *f, unsigned gx,
+					    unsigned gy, unsigned gz,
+					    unsigned bx, unsigned by,
+					    unsigned bz, unsigned shmem,
+					    hipStream_t s, void **params)
+{
+	(void)f; (void)gx; (void)gy; (void)gz; (void)bx; (void)by;
+	(void)bz; (void)shmem; (void)s; (void)params;
+	return run_kernel();
+}
+
 hipError_t hipMemset(void *d, int v, size_t n)
 {
 	ev("memset", (long long)n);
