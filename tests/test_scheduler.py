@@ -298,3 +298,41 @@ def test_per_gpu_tq_override(sched, sock_dir):
         a0.recv(2.0)
     for c in (a0, b0, a1, b1):
         c.close()
+
+
+def test_round_robin_rotation(sched, sock_dir):
+    """Three clients that re-request after each release rotate in FCFS
+    order across multiple quanta (A, B, C, A, B, ...)."""
+    clients = {n: make_client(sock_dir, n) for n in "abc"}
+    order = []
+    for n in "abc":
+        clients[n].send(proto.REQ_LOCK)
+        time.sleep(0.05)
+    for _ in range(6):
+        granted = None
+        for n, c in clients.items():
+            try:
+                m = c.recv(0.1)
+            except Exception:
+                continue
+            if m.type == proto.LOCK_OK:
+                granted = n
+                break
+        if granted is None:
+            # wait for the next grant on any socket
+            for n, c in clients.items():
+                try:
+                    m = c.recv(3)
+                except Exception:
+                    continue
+                if m.type == proto.LOCK_OK:
+                    granted = n
+                    break
+        assert granted is not None
+        order.append(granted)
+        clients[granted].send(proto.LOCK_RELEASED)
+        clients[granted].send(proto.REQ_LOCK)  # back of the queue
+    # Strict FCFS: the rotation must cycle a,b,c,a,b,c.
+    assert order == list("abcabc"), order
+    for c in clients.values():
+        c.close()
