@@ -1,6 +1,7 @@
 /* nvshare-amd wire protocol implementation. */
 #define _GNU_SOURCE
 #include <fcntl.h>
+#include <poll.h>
 #include <stdio.h>
 #include <string.h>
 #include <sys/socket.h>
@@ -129,10 +130,45 @@ int nvs_accept(int lsock)
 	return fd;
 }
 
+/*
+ * Frame-safe send that also works on nonblocking fds (the daemon keeps
+ * client sockets nonblocking for reads): on EAGAIN, poll for
+ * writability briefly (500 ms total) so a frame is never torn
+ * mid-message.  The budget is short on purpose: the daemon sends while
+ * holding its global lock, so a slow peer must not head-of-line block
+ * other clients (the reference evicted instantly on partial sends; we
+ * just add a small grace).  A peer that stays unwritable is broken and
+ * the caller must treat the failure as fatal for the connection
+ * (framing can no longer be trusted).
+ */
 int nvs_send_msg(int fd, const struct nvs_msg *m)
 {
-	if (nvs_write_whole(fd, m, NVS_MSG_SIZE) != NVS_MSG_SIZE)
+	const char *p = (const char *)m;
+	size_t left = NVS_MSG_SIZE;
+	int waited_ms = 0;
+
+	while (left > 0) {
+		ssize_t n;
+
+		RETRY_EINTR(n, write(fd, p, left));
+		if (n > 0) {
+			p += n;
+			left -= (size_t)n;
+			continue;
+		}
+		if (n < 0 && (errno == EAGAIN || errno == EWOULDBLOCK)) {
+			struct pollfd pfd = { .fd = fd,
+					      .events = POLLOUT };
+			int pr;
+
+			if (waited_ms >= 500)
+				return -1;
+			RETRY_EINTR(pr, poll(&pfd, 1, 50));
+			waited_ms += 50;
+			continue;
+		}
 		return -1;
+	}
 	return 0;
 }
 

@@ -59,6 +59,7 @@ struct conn {
 	char pod_namespace[NVS_POD_NS_LEN];
 	int wants_lock;           /* present in its GPU's request queue */
 	int relock_pending;       /* holder re-requested during release */
+	int dead;                 /* unwritable; evict at a safe point */
 	long mem_mib;             /* client-reported tracked allocations */
 	struct conn *next;        /* registry list */
 	struct conn *qnext;       /* FCFS queue list */
@@ -125,11 +126,19 @@ static void send_to(struct conn *c, uint8_t type, const char *data)
 {
 	struct nvs_msg m;
 
+	if (c->dead)
+		return; /* already unwritable; sweep will evict it */
 	nvs_msg_init(&m, type, c->id, data);
-	if (nvs_send_msg(c->fd, &m) != 0)
-		log_warn("send %s to client %016" PRIx64 " failed",
-			 nvs_msg_type_str(type), c->id);
-	else
+	if (nvs_send_msg(c->fd, &m) != 0) {
+		/* Framing toward this peer can no longer be trusted:
+		 * mark it for eviction at the end of the event-loop
+		 * pass (deleting here would invalidate pointers the
+		 * callers still hold). */
+		log_warn("send %s to client %016" PRIx64 " failed; "
+			 "marking for eviction", nvs_msg_type_str(type),
+			 c->id);
+		c->dead = 1;
+	} else
 		log_debug("sent %s to client %016" PRIx64,
 			  nvs_msg_type_str(type), c->id);
 }
@@ -527,6 +536,8 @@ static int handle_readable(struct conn *c)
 		if (c->got == NVS_MSG_SIZE) {
 			c->got = 0;
 			process_msg(c, &c->inmsg);
+			if (c->dead)
+				return 0; /* stop; sweep evicts it */
 		}
 	}
 }
@@ -672,6 +683,22 @@ int main(void)
 			if (events[i].events & EPOLLIN) {
 				if (handle_readable(c) < 0)
 					delete_conn(c);
+			}
+		}
+		/* Sweep connections whose sends failed this pass. */
+		{
+			struct conn *c;
+			int again = 1;
+
+			while (again) {
+				again = 0;
+				for (c = clients; c != NULL; c = c->next) {
+					if (c->dead) {
+						delete_conn(c);
+						again = 1;
+						break;
+					}
+				}
 			}
 		}
 		pthread_mutex_unlock(&g_mutex);

@@ -67,13 +67,17 @@ def test_unknown_message_types(sched, sock_dir):
 
 
 def test_flooding_client(sched, sock_dir):
-    """A client spamming REQ_LOCK/RELEASED cannot wedge the daemon."""
+    """A client spamming REQ_LOCK/RELEASED without reading replies
+    cannot wedge the daemon; the daemon may evict it (strict eviction,
+    like the reference on partial sends) but must keep serving."""
     c = proto.Client(sock_dir=sock_dir, pod_name="flood").connect()
     c.register()
-    for _ in range(500):
-        c.send(proto.REQ_LOCK)
-        c.send(proto.LOCK_RELEASED)
-    # Drain whatever grants arrived, then confirm liveness.
+    try:
+        for _ in range(500):
+            c.send(proto.REQ_LOCK)
+            c.send(proto.LOCK_RELEASED)
+    except (BrokenPipeError, ConnectionError):
+        pass  # evicted mid-flood: acceptable
     try:
         while True:
             c.recv(0.5)
@@ -81,6 +85,12 @@ def test_flooding_client(sched, sock_dir):
         pass
     assert alive(sock_dir)
     c.close()
+    # A well-behaved client still gets the full lifecycle afterwards.
+    ok = proto.Client(sock_dir=sock_dir, pod_name="after").connect()
+    ok.register()
+    ok.send(proto.REQ_LOCK)
+    assert ok.recv(5).type == proto.LOCK_OK
+    ok.close()
 
 
 def test_many_connections(sched, sock_dir):
