@@ -502,11 +502,13 @@ void nvs_prefetch_allocs(void)
 
 /*
  * Evict tracked allocations to host after giving up the lock
- * (NVSHARE_EVICT=1, called from client.c after LOCK_RELEASED).
+ * (called from client.c BEFORE LOCK_RELEASED; on by default via
+ * NVSHARE_AUTO_MIGRATE, self-gated on real memory pressure below).
  * Explicit eviction runs at ~10.9 GB/s on MI355X vs ~0.15 GB/s when
  * the next client's demand faults push pages out one at a time
- * (profiles/restorebench.json).  Fire-and-forget on the prefetch
- * streams: it overlaps the next client's quantum.
+ * (profiles/restorebench.json).  Blocking: the next client is granted
+ * only once the room actually exists, otherwise its restore races the
+ * driver's fault-driven eviction of our pages.
  */
 void nvs_evict_allocs(void)
 {
