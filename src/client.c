@@ -477,6 +477,12 @@ static void *early_release_thread(void *arg)
 		own_lock = 0;
 		pthread_mutex_unlock(&g_mutex);
 		drain_gpu();
+		/* Same pressure-gated eviction as the DROP_LOCK path:
+		 * leaving the idle set resident makes the next holder's
+		 * restore fight fault-driven eviction. */
+		if (nvs_env_bool("NVSHARE_EVICT",
+				 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
+			nvs_evict_allocs();
 		send_msg_type(NVS_LOCK_RELEASED);
 		log_debug("client: early-released idle lock");
 	}

@@ -83,6 +83,8 @@ struct nvs_alloc {
 	void *ptr;
 	size_t size;
 	int populated;   /* device pages materialized (eager prefetch) */
+	int passthrough; /* real-VRAM (stream-ordered) alloc: counted
+			  * against the cap but never migrated */
 	struct nvs_alloc *next;
 };
 static struct nvs_alloc *alloc_list;
@@ -313,7 +315,12 @@ static int populate_managed(void *ptr, size_t size)
 	return 1;
 }
 
-static void track_alloc(void *ptr, size_t size, int populated)
+/* reserved=1: the size was already added to sum_allocated by the
+ * caller (cap reservation under alloc_mutex before the real alloc —
+ * check-then-act otherwise lets concurrent allocations jointly
+ * exceed the cap). */
+static void track_alloc2(void *ptr, size_t size, int populated,
+			 int reserved, int passthrough)
 {
 	struct nvs_alloc *a = malloc(sizeof(*a));
 
@@ -322,16 +329,53 @@ static void track_alloc(void *ptr, size_t size, int populated)
 	a->ptr = ptr;
 	a->size = size;
 	a->populated = populated;
+	a->passthrough = passthrough;
 	pthread_mutex_lock(&alloc_mutex);
 	a->next = alloc_list;
 	alloc_list = a;
-	sum_allocated += size;
+	if (!reserved)
+		sum_allocated += size;
 	if (!populated)
 		__atomic_fetch_add(&n_unpopulated, 1, __ATOMIC_RELAXED);
 	pthread_mutex_unlock(&alloc_mutex);
-	log_debug("hook: +alloc %p %zu MiB (sum %zu MiB%s)", ptr,
+	log_debug("hook: +alloc %p %zu MiB (sum %zu MiB%s%s)", ptr,
 		  size / NVS_MIB, sum_allocated / NVS_MIB,
-		  populated ? "" : ", population deferred");
+		  populated ? "" : ", population deferred",
+		  passthrough ? ", passthrough" : "");
+}
+
+static void track_alloc(void *ptr, size_t size, int populated,
+			int reserved)
+{
+	track_alloc2(ptr, size, populated, reserved, 0);
+}
+
+/* Reserve size against the cap (sum_allocated) before allocating.
+ * Returns 0 on success, -1 when the cap would be exceeded. */
+static int reserve_cap(size_t size)
+{
+	pthread_mutex_lock(&alloc_mutex);
+	if (sum_allocated + size > mem_limit()) {
+		size_t sum = sum_allocated;
+
+		pthread_mutex_unlock(&alloc_mutex);
+		log_debug("hook: reject alloc of %zu MiB (sum %zu MiB, "
+			  "limit %zu MiB); set "
+			  "NVSHARE_ENABLE_SINGLE_OVERSUB=1 to oversubscribe",
+			  size / NVS_MIB, sum / NVS_MIB,
+			  mem_limit() / NVS_MIB);
+		return -1;
+	}
+	sum_allocated += size;
+	pthread_mutex_unlock(&alloc_mutex);
+	return 0;
+}
+
+static void unreserve_cap(size_t size)
+{
+	pthread_mutex_lock(&alloc_mutex);
+	sum_allocated -= size;
+	pthread_mutex_unlock(&alloc_mutex);
 }
 
 long nvs_sum_allocated_mib(void)
@@ -370,12 +414,15 @@ void nvs_populate_pending(void)
 		log_debug("hook: populated deferred ranges");
 }
 
-/* Returns tracked size, or 0 if unknown pointer. */
-static size_t untrack_alloc(void *ptr)
+/* Returns tracked size, or 0 if unknown pointer.  was_passthrough
+ * (optional) reports whether the range was a real-VRAM passthrough. */
+static size_t untrack_alloc2(void *ptr, int *was_passthrough)
 {
 	struct nvs_alloc **pp, *a;
 	size_t size = 0;
 
+	if (was_passthrough != NULL)
+		*was_passthrough = 0;
 	pthread_mutex_lock(&alloc_mutex);
 	for (pp = &alloc_list; *pp != NULL; pp = &(*pp)->next) {
 		if ((*pp)->ptr == ptr) {
@@ -383,6 +430,8 @@ static size_t untrack_alloc(void *ptr)
 			*pp = a->next;
 			size = a->size;
 			sum_allocated -= size;
+			if (was_passthrough != NULL)
+				*was_passthrough = a->passthrough;
 			if (!a->populated)
 				__atomic_fetch_sub(&n_unpopulated, 1,
 						   __ATOMIC_RELAXED);
@@ -395,6 +444,11 @@ static size_t untrack_alloc(void *ptr)
 		log_debug("hook: -alloc %p %zu MiB (sum %zu MiB)", ptr,
 			  size / NVS_MIB, sum_allocated / NVS_MIB);
 	return size;
+}
+
+static size_t untrack_alloc(void *ptr)
+{
+	return untrack_alloc2(ptr, NULL);
 }
 
 /*
@@ -473,6 +527,8 @@ void nvs_prefetch_allocs(void)
 		size_t left = a->size < budget ? a->size : budget;
 		char *p = a->ptr;
 
+		if (a->passthrough)
+			continue; /* real VRAM: never migrated */
 		if (!a->populated) {
 			a->populated = 1;
 			__atomic_fetch_sub(&n_unpopulated, 1,
@@ -537,6 +593,8 @@ void nvs_evict_allocs(void)
 		char *p = a->ptr;
 		size_t left = a->size;
 
+		if (a->passthrough)
+			continue; /* real VRAM: never migrated */
 		while (left > 0) {
 			size_t n = left < PREFETCH_CHUNK ? left :
 				   PREFETCH_CHUNK;
@@ -709,12 +767,12 @@ __attribute__((destructor)) static void dump_hook_counts(void)
 /* Hooked entry points                                                 */
 /* ------------------------------------------------------------------ */
 
-nvshipError_t hipMalloc(void **ptr, size_t size)
+/* Shared managed-conversion path.  Counters are bumped only at the
+ * public entry points so delegating wrappers don't double-count. */
+static nvshipError_t malloc_managed(void **ptr, size_t size)
 {
 	nvshipError_t r;
 
-	BOOTSTRAP();
-	BUMP(H_hipMalloc);
 	CHECK_REAL(hipMalloc);
 	if (ptr == NULL)
 		return NVSHIP_ERROR_INVALID_VALUE;
@@ -722,24 +780,22 @@ nvshipError_t hipMalloc(void **ptr, size_t size)
 		return real.hipMalloc(ptr, size);
 	if (disable_um)
 		return real.hipMalloc(ptr, size);
-	if (!oversub_allowed) {
-		pthread_mutex_lock(&alloc_mutex);
-		if (sum_allocated + size > mem_limit()) {
-			pthread_mutex_unlock(&alloc_mutex);
-			log_debug("hook: reject alloc of %zu MiB "
-				  "(sum %zu MiB, limit %zu MiB); set "
-				  "NVSHARE_ENABLE_SINGLE_OVERSUB=1 to "
-				  "oversubscribe", size / NVS_MIB,
-				  sum_allocated / NVS_MIB,
-				  mem_limit() / NVS_MIB);
-			return NVSHIP_ERROR_OOM;
-		}
-		pthread_mutex_unlock(&alloc_mutex);
-	}
+	if (!oversub_allowed && reserve_cap(size) != 0)
+		return NVSHIP_ERROR_OOM;
 	r = real.hipMallocManaged(ptr, size, NVSHIP_MEM_ATTACH_GLOBAL);
 	if (r == NVSHIP_SUCCESS)
-		track_alloc(*ptr, size, populate_managed(*ptr, size));
+		track_alloc(*ptr, size, populate_managed(*ptr, size),
+			    !oversub_allowed);
+	else if (!oversub_allowed)
+		unreserve_cap(size);
 	return r;
+}
+
+nvshipError_t hipMalloc(void **ptr, size_t size)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMalloc);
+	return malloc_managed(ptr, size);
 }
 
 nvshipError_t hipExtMallocWithFlags(void **ptr, size_t size,
@@ -751,7 +807,7 @@ nvshipError_t hipExtMallocWithFlags(void **ptr, size_t size,
 		return real.hipExtMallocWithFlags(ptr, size, flags);
 	/* All flag variants become managed; the flags are advisory. */
 	(void)flags;
-	return hipMalloc(ptr, size);
+	return malloc_managed(ptr, size);
 }
 
 /*
@@ -774,18 +830,57 @@ static int pool_passthrough(size_t size)
 	return !disable_um && size <= (size_t)thresh_mib * NVS_MIB;
 }
 
+/* Passthrough allocations stay real VRAM but are still COUNTED: they
+ * reserve against the cap, appear in hipMemGetInfo accounting and in
+ * MEM_UPDATE, and are untracked on free (the reference counts every
+ * byte, hook.c:662-670; a client must not be able to dodge the cap by
+ * allocating its whole set in small pool chunks). */
+static nvshipError_t pool_alloc_passthrough(
+	void **ptr, size_t size,
+	nvshipError_t (*do_alloc)(void **, size_t))
+{
+	nvshipError_t r;
+
+	if (!disable_um && !oversub_allowed && size > 0 &&
+	    reserve_cap(size) != 0)
+		return NVSHIP_ERROR_OOM;
+	r = do_alloc(ptr, size);
+	if (r == NVSHIP_SUCCESS) {
+		if (!disable_um && size > 0)
+			track_alloc2(*ptr, size, 1, !oversub_allowed, 1);
+	} else if (!disable_um && !oversub_allowed && size > 0) {
+		unreserve_cap(size);
+	}
+	return r;
+}
+
+/* Thunks adapting stream/pool args for pool_alloc_passthrough. */
+static __thread nvship_stream_t pp_stream;
+static __thread nvship_mempool_t pp_pool;
+
+static nvshipError_t do_malloc_async(void **ptr, size_t size)
+{
+	return real.hipMallocAsync(ptr, size, pp_stream);
+}
+
+static nvshipError_t do_malloc_from_pool_async(void **ptr, size_t size)
+{
+	return real.hipMallocFromPoolAsync(ptr, size, pp_pool, pp_stream);
+}
+
 nvshipError_t hipMallocAsync(void **ptr, size_t size, nvship_stream_t s)
 {
 	BOOTSTRAP();
 	BUMP(H_hipMallocAsync);
-	if (disable_um ||
-	    (pool_passthrough(size) && real.hipMallocAsync != NULL)) {
-		CHECK_REAL(hipMallocAsync);
+	if (disable_um && real.hipMallocAsync != NULL)
 		return real.hipMallocAsync(ptr, size, s);
+	if (pool_passthrough(size) && real.hipMallocAsync != NULL) {
+		pp_stream = s;
+		return pool_alloc_passthrough(ptr, size, do_malloc_async);
 	}
 	/* Large stream-ordered alloc becomes an immediate managed alloc:
 	 * the pointer is valid earlier than required (safe). */
-	return hipMalloc(ptr, size);
+	return malloc_managed(ptr, size);
 }
 
 nvshipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
@@ -794,20 +889,21 @@ nvshipError_t hipMallocFromPoolAsync(void **ptr, size_t size,
 {
 	BOOTSTRAP();
 	BUMP(H_hipMallocFromPoolAsync);
-	if (disable_um ||
-	    (pool_passthrough(size) &&
-	     real.hipMallocFromPoolAsync != NULL)) {
-		CHECK_REAL(hipMallocFromPoolAsync);
+	if (disable_um && real.hipMallocFromPoolAsync != NULL)
 		return real.hipMallocFromPoolAsync(ptr, size, pool, s);
+	if (pool_passthrough(size) && real.hipMallocFromPoolAsync != NULL) {
+		pp_stream = s;
+		pp_pool = pool;
+		return pool_alloc_passthrough(ptr, size,
+					      do_malloc_from_pool_async);
 	}
-	return hipMalloc(ptr, size);
+	return malloc_managed(ptr, size);
 }
 
 nvshipError_t hipFree(void *ptr)
 {
 	BOOTSTRAP();
 	BUMP(H_hipFree);
-	BUMP(H_hipMallocFromPoolAsync);
 	CHECK_REAL(hipFree);
 	if (ptr != NULL)
 		untrack_alloc(ptr);
@@ -816,11 +912,13 @@ nvshipError_t hipFree(void *ptr)
 
 nvshipError_t hipFreeAsync(void *ptr, nvship_stream_t stream)
 {
+	int was_pt = 0;
+
 	BOOTSTRAP();
 	BUMP(H_hipFreeAsync);
 	if (ptr == NULL)
 		return NVSHIP_SUCCESS;
-	if (untrack_alloc(ptr) != 0) {
+	if (untrack_alloc2(ptr, &was_pt) != 0 && !was_pt) {
 		/* We converted this to a managed alloc: stream-ordered
 		 * free semantics require prior stream work to finish. */
 		CHECK_REAL(hipFree);
@@ -1143,23 +1241,25 @@ nvshipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags)
 	CHECK_REAL(hipMallocManaged);
 	if (ptr == NULL)
 		return NVSHIP_ERROR_INVALID_VALUE;
-	if (!disable_um && !oversub_allowed && size > 0) {
-		pthread_mutex_lock(&alloc_mutex);
-		if (sum_allocated + size > mem_limit()) {
-			pthread_mutex_unlock(&alloc_mutex);
-			return NVSHIP_ERROR_OOM;
-		}
-		pthread_mutex_unlock(&alloc_mutex);
-	}
+	if (!disable_um && !oversub_allowed && size > 0 &&
+	    reserve_cap(size) != 0)
+		return NVSHIP_ERROR_OOM;
 	r = real.hipMallocManaged(ptr, size, flags);
 	if (r == NVSHIP_SUCCESS && size > 0 && !disable_um)
-		track_alloc(*ptr, size, 1); /* app-managed: don't touch */
+		track_alloc(*ptr, size, 1, !oversub_allowed);
+	else if (r != NVSHIP_SUCCESS && !disable_um && !oversub_allowed &&
+		 size > 0)
+		unreserve_cap(size);
 	return r;
 }
 
 /* ------------------------------------------------------------------ */
 /* Entry-point rerouting: hipGetProcAddress + versioned dlsym           */
 /* ------------------------------------------------------------------ */
+
+nvshipError_t hipGetProcAddress(const char *symbol, void **pfn,
+				int hip_version, uint64_t flags,
+				void *symbol_status);
 
 struct hook_entry {
 	const char *name;
@@ -1202,6 +1302,11 @@ static const struct hook_entry hook_table[] = {
 	{ "hipModuleLaunchCooperativeKernel",
 	  (void *)hipModuleLaunchCooperativeKernel },
 	{ "hipMallocManaged", (void *)hipMallocManaged },
+	/* dlsym-resolved hipGetProcAddress must return OUR wrapper,
+	 * else every entry point fetched through it silently bypasses
+	 * the cap and the gate (the reference hooked its
+	 * cuGetProcAddress equivalent in the dlsym path too). */
+	{ "hipGetProcAddress", (void *)hipGetProcAddress },
 	{ NULL, NULL },
 };
 

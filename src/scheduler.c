@@ -320,7 +320,6 @@ static long total_mem_mib(void)
 
 static void handle_status_req(struct conn *c)
 {
-	struct nvs_msg m;
 	char big[64], buf[NVS_MSG_DATA_LEN];
 
 	/* data: "<on>,<tq>,<nclients>,<qlen>[,<mem_mib>]" — the wire
@@ -329,9 +328,10 @@ static void handle_status_req(struct conn *c)
 		 tq_seconds, client_count(), queue_len_all(),
 		 total_mem_mib());
 	nvs_strlcpy(buf, big, sizeof(buf));
-	nvs_msg_init(&m, NVS_STATUS, c->id, buf);
-	if (nvs_send_msg(c->fd, &m) != 0)
-		log_warn("STATUS reply failed");
+	/* Through send_to so a failed/partial write marks the peer dead
+	 * and the sweep evicts it — a torn frame on a kept-alive fd
+	 * would misframe every later send to that peer. */
+	send_to(c, NVS_STATUS, buf);
 }
 
 static int parse_gpu_index(const struct nvs_msg *m)

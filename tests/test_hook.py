@@ -205,6 +205,44 @@ int main(void) {
     assert "hipMallocManaged" in names
 
 
+def test_pool_passthrough_counted_against_cap(artifacts, sched, sock_dir):
+    """Passthrough (small stream-ordered) allocations still reserve
+    against the cap: a client allocating its whole set in <=threshold
+    chunks cannot dodge the limit (reference invariant: every byte
+    counted, hook.c:662-670)."""
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     reserve_mib=16, fake_total_mib=64)
+    env["NVSTUB_TOTAL_MIB"] = "4096"
+    env["NVSHARE_PASSTHROUGH_MIB"] = "8"
+    # limit = 64 - 16 = 48 MiB -> at most 12 chunks of 4 MiB.
+    code = r"""
+#include <stdio.h>
+typedef int hipError_t;
+extern hipError_t hipMallocAsync(void **, unsigned long, void *);
+int main(void) {
+    void *p; int n = 0;
+    while (n < 100 && hipMallocAsync(&p, 4ul << 20, 0) == 0)
+        n++;
+    printf("CHUNKS=%d\n", n);
+    return 0;
+}
+"""
+    src = os.path.join(sock_dir, "cap.c")
+    exe = os.path.join(sock_dir, "cap")
+    with open(src, "w") as f:
+        f.write(code)
+    build = subprocess.run(
+        ["gcc", "-o", exe, src, "-L", str(artifacts.stub_dir),
+         "-lamdhip64", f"-Wl,-rpath,{artifacts.stub_dir}"],
+        capture_output=True, text=True)
+    assert build.returncode == 0, build.stderr
+    r = subprocess.run([exe], env=env, capture_output=True, text=True,
+                       timeout=60)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    chunks = int(r.stdout.split("CHUNKS=")[1].split()[0])
+    assert chunks == 12, (chunks, r.stderr[-2000:])
+
+
 def test_fork_safety(artifacts, sched, sock_dir):
     """A forked child free-runs without corrupting the parent's
     scheduler protocol or deadlocking (reference would deadlock)."""
