@@ -53,6 +53,22 @@ static int need_lock = 0;
 static int did_work = 0;
 static int standalone = 0;
 static uint64_t client_id = 0;
+/* Scheduler's global memory-pressure verdict, from LOCK_OK "p=" data:
+ * -1 unknown, 0 fits, 1 oversubscribed.  Drives migration assist when
+ * NVSHARE_PREFETCH/NVSHARE_EVICT are not explicitly set. */
+static int sched_pressure = -1;
+
+/* Migration-assist policy: an explicit env setting wins; else the
+ * scheduler's pressure verdict; else the AUTO_MIGRATE default (which
+ * self-gates on real free memory in hook.c). */
+static int migrate_enabled(const char *env)
+{
+	if (getenv(env) != NULL)
+		return nvs_env_bool(env, 1);
+	if (__atomic_load_n(&sched_pressure, __ATOMIC_RELAXED) >= 0)
+		return __atomic_load_n(&sched_pressure, __ATOMIC_RELAXED);
+	return nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1);
+}
 
 /* Submission lock: readers = app threads inside a real HIP call,
  * writer = the drain path (DROP_LOCK / early release). */
@@ -259,16 +275,22 @@ static int connect_and_register(void)
 	return fd;
 }
 
-static void handle_lock_ok(void)
+static void handle_lock_ok(const struct nvs_msg *m)
 {
+	/* The scheduler includes its whole-node pressure verdict with
+	 * the grant ("p=0"/"p=1"); reference-era schedulers send no
+	 * data, leaving the verdict unknown. */
+	if (m->data[0] == 'p' && m->data[1] == '=')
+		__atomic_store_n(&sched_pressure, m->data[2] == '1' ? 1 : 0,
+				 __ATOMIC_RELAXED);
 	/* Restore the working set BEFORE opening the gate: demand
-	 * faults racing the bulk migration degrade both.  Default
-	 * (NVSHARE_AUTO_MIGRATE=1): always restore on grant — when the
-	 * set is already resident the prefetch is a no-op costing
-	 * microseconds; when it was evicted this is 25x faster than
-	 * demand refaulting (profiles/restorebench.json). */
-	if (nvs_env_bool("NVSHARE_PREFETCH",
-			 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
+	 * faults racing the bulk migration degrade both.  Under
+	 * pressure (or with NVSHARE_PREFETCH=1 / the AUTO_MIGRATE
+	 * fallback) restore on grant — when the set is already resident
+	 * the prefetch is a no-op costing microseconds; when it was
+	 * evicted this is 25x faster than demand refaulting
+	 * (profiles/restorebench.json). */
+	if (migrate_enabled("NVSHARE_PREFETCH"))
 		nvs_prefetch_allocs();
 	pthread_mutex_lock(&g_mutex);
 	own_lock = 1;
@@ -288,11 +310,10 @@ static void handle_drop_lock(void)
 	if (!had)
 		return; /* already released voluntarily */
 	drain_gpu();
-	/* Eviction self-gates on real memory pressure (it is skipped
-	 * when free HBM already fits the tracked set), so the automatic
-	 * default is safe for fitting workloads. */
-	if (nvs_env_bool("NVSHARE_EVICT",
-			 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
+	/* Eviction additionally self-gates on real memory pressure (it
+	 * is skipped when free HBM already fits the tracked set), so the
+	 * automatic default is safe for fitting workloads. */
+	if (migrate_enabled("NVSHARE_EVICT"))
 		nvs_evict_allocs(); /* blocking; under pressure only */
 	send_msg_type(NVS_LOCK_RELEASED);
 	log_debug("client: lock released after DROP_LOCK");
@@ -358,7 +379,7 @@ static void *client_thread(void *arg)
 		}
 		switch (m.type) {
 		case NVS_LOCK_OK:
-			handle_lock_ok();
+			handle_lock_ok(&m);
 			break;
 		case NVS_DROP_LOCK:
 			handle_drop_lock();
@@ -451,7 +472,11 @@ static void *early_release_thread(void *arg)
 			struct nvs_msg m;
 			char buf[NVS_MSG_DATA_LEN];
 
-			snprintf(buf, sizeof(buf), "%ld", mem_mib);
+			/* "mem,cap": capacity feeds the scheduler's
+			 * pressure policy.  The 20-byte data field fits
+			 * both for any real MiB value. */
+			snprintf(buf, sizeof(buf), "%ld,%ld", mem_mib,
+				 nvs_mem_total_mib());
 			nvs_msg_init(&m, NVS_MEM_UPDATE, client_id, buf);
 			pthread_mutex_lock(&sock_mutex);
 			if (sock_fd >= 0 &&
@@ -480,8 +505,7 @@ static void *early_release_thread(void *arg)
 		/* Same pressure-gated eviction as the DROP_LOCK path:
 		 * leaving the idle set resident makes the next holder's
 		 * restore fight fault-driven eviction. */
-		if (nvs_env_bool("NVSHARE_EVICT",
-				 nvs_env_bool("NVSHARE_AUTO_MIGRATE", 1)))
+		if (migrate_enabled("NVSHARE_EVICT"))
 			nvs_evict_allocs();
 		send_msg_type(NVS_LOCK_RELEASED);
 		log_debug("client: early-released idle lock");

@@ -96,4 +96,7 @@ void nvs_evict_allocs(void);
 /* Sum of tracked allocations in MiB (hook.c). */
 long nvs_sum_allocated_mib(void);
 
+/* Advertised device capacity in MiB (hook.c). */
+long nvs_mem_total_mib(void);
+
 #endif /* NVSHARE_CLIENT_H */

@@ -388,6 +388,14 @@ long nvs_sum_allocated_mib(void)
 	return (long)(n / NVS_MIB);
 }
 
+/* Advertised device capacity (MiB), for MEM_UPDATE so the scheduler
+ * can judge whole-node memory pressure. */
+long nvs_mem_total_mib(void)
+{
+	return (long)(mem_limit() / NVS_MIB) +
+	       (long)(mem_reserve / NVS_MIB);
+}
+
 /* Called from the gate with the submission read lock held and the GPU
  * lock owned: materialize any deferred ranges before real work. */
 void nvs_populate_pending(void)

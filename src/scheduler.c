@@ -61,6 +61,7 @@ struct conn {
 	int relock_pending;       /* holder re-requested during release */
 	int dead;                 /* unwritable; evict at a safe point */
 	long mem_mib;             /* client-reported tracked allocations */
+	long cap_mib;             /* client-reported device capacity */
 	struct conn *next;        /* registry list */
 	struct conn *qnext;       /* FCFS queue list */
 };
@@ -172,9 +173,39 @@ static void queue_remove(struct gpu_state *g, struct conn *c)
 	c->wants_lock = 0;
 }
 
+/*
+ * Global memory-pressure verdict for one GPU: 1 when the clients'
+ * combined tracked working sets exceed the device capacity they
+ * report, 0 when they fit, -1 when no client has reported a capacity
+ * yet.  Broadcast with LOCK_OK ("p=0"/"p=1") so clients flip their
+ * migration assist (evict-on-release / prefetch-on-grant) on or off
+ * from the scheduler's whole-node view instead of per-process env
+ * guesses (docs/roadmap.md #2).
+ */
+static int gpu_pressure(const struct gpu_state *g)
+{
+	const struct conn *c;
+	long sum = 0, cap = 0;
+	int gi = (int)(g - gpus);
+
+	for (c = clients; c != NULL; c = c->next) {
+		if (!c->registered || c->gpu != gi)
+			continue;
+		sum += c->mem_mib;
+		if (c->cap_mib > cap)
+			cap = c->cap_mib;
+	}
+	if (cap <= 0)
+		return -1;
+	return sum > cap;
+}
+
 /* Grant GPU g's lock to its queue head if possible. Holds g_mutex. */
 static void try_schedule(struct gpu_state *g)
 {
+	int pressure;
+	const char *pdata;
+
 	if (!scheduler_on || g->lock_held || g->queue_head == NULL)
 		return;
 	g->lock_holder = g->queue_head;
@@ -183,7 +214,9 @@ static void try_schedule(struct gpu_state *g)
 	g->round++;
 	g->grants++;
 	g->quantum_start_ns = nvs_now_ns();
-	send_to(g->lock_holder, NVS_LOCK_OK, NULL);
+	pressure = gpu_pressure(g);
+	pdata = pressure < 0 ? NULL : (pressure ? "p=1" : "p=0");
+	send_to(g->lock_holder, NVS_LOCK_OK, pdata);
 	log_debug("gpu%ld round %lu: lock -> %016" PRIx64 " (queue=%d)",
 		  (long)(g - gpus), g->round, g->lock_holder->id,
 		  queue_len(g));
@@ -496,13 +529,22 @@ static void process_msg(struct conn *c, const struct nvs_msg *m)
 		break;
 	case NVS_MEM_UPDATE: {
 		char buf[NVS_MSG_DATA_LEN];
+		char *end = NULL;
 		long v;
 
 		memcpy(buf, m->data, NVS_MSG_DATA_LEN);
 		buf[NVS_MSG_DATA_LEN - 1] = '\0';
-		v = strtol(buf, NULL, 10);
+		/* "mem_mib[,cap_mib]" — cap feeds the global pressure
+		 * policy (see gpu_pressure). */
+		v = strtol(buf, &end, 10);
 		if (c->registered && v >= 0)
 			c->mem_mib = v;
+		if (c->registered && end != NULL && *end == ',') {
+			long cap = strtol(end + 1, NULL, 10);
+
+			if (cap > 0)
+				c->cap_mib = cap;
+		}
 		break;
 	}
 	case NVS_STATUS_REQ:

@@ -212,6 +212,43 @@ def test_mem_update_reported_in_status(sched, sock_dir):
     b.close()
 
 
+def test_pressure_flag_in_lock_ok(sched, sock_dir):
+    """LOCK_OK carries the scheduler's whole-node pressure verdict:
+    p=1 when the clients' combined tracked sets exceed the reported
+    device capacity, p=0 when they fit, no data when no client has
+    reported a capacity (reference-era clients)."""
+    a = make_client(sock_dir, "a")
+    b = make_client(sock_dir, "b")
+
+    # No capacity reported yet: grant carries no verdict.
+    a.send(proto.REQ_LOCK)
+    m = a.recv(5)
+    assert m.type == proto.LOCK_OK
+    assert m.data == ""
+    a.send(proto.LOCK_RELEASED)
+
+    # Fits: 1024 + 512 < 4096.
+    a.send(proto.MEM_UPDATE, "1024,4096")
+    b.send(proto.MEM_UPDATE, "512,4096")
+    time.sleep(0.2)
+    a.send(proto.REQ_LOCK)
+    m = a.recv(5)
+    assert m.type == proto.LOCK_OK
+    assert m.data == "p=0"
+    a.send(proto.LOCK_RELEASED)
+
+    # Oversubscribed: 3000 + 2000 > 4096.
+    a.send(proto.MEM_UPDATE, "3000,4096")
+    b.send(proto.MEM_UPDATE, "2000,4096")
+    time.sleep(0.2)
+    b.send(proto.REQ_LOCK)
+    m = b.recv(5)
+    assert m.type == proto.LOCK_OK
+    assert m.data == "p=1"
+    a.close()
+    b.close()
+
+
 def test_stale_lock_released_ignored(sched, sock_dir):
     a = make_client(sock_dir, "a")
     b = make_client(sock_dir, "b")
