@@ -38,6 +38,27 @@ def client_env(
     env["LD_PRELOAD"] = preload
     env["HSA_XNACK"] = "1"
 
+    # PyTorch's expandable-segments allocator backend goes through
+    # hipMemCreate/hipMemMap: the interposer caps it (hook.c) but the
+    # physical VMM handles cannot be converted to managed memory, so
+    # oversubscription/migration would silently not apply.  Strip the
+    # option so clients use the default (hipMalloc) backend.
+    for var in ("PYTORCH_HIP_ALLOC_CONF", "PYTORCH_CUDA_ALLOC_CONF"):
+        conf = env.get(var)
+        if conf and "expandable_segments" in conf:
+            kept = [p for p in conf.split(",")
+                    if "expandable_segments" not in p]
+            if kept:
+                env[var] = ",".join(kept)
+            else:
+                env.pop(var, None)
+
+    # Deterministic MIOpen conv-algo selection: heuristic-based FAST
+    # find instead of a cold exhaustive search (a fresh box otherwise
+    # runs ~50k tuning kernels inside the gated region, round-1
+    # GPUTEST fresh-box hang).  Callers can override via base/extra.
+    env.setdefault("MIOPEN_FIND_MODE", "FAST")
+
     if sock_dir:
         env["NVSHARE_SOCK_DIR"] = sock_dir
     env["NVSHARE_DEBUG"] = "1" if debug else env.get("NVSHARE_DEBUG", "0")

@@ -57,6 +57,10 @@ struct nvs_real_hip {
 	fn_hipModuleLaunchCooperativeKernel hipModuleLaunchCooperativeKernel;
 	fn_hipGetProcAddress hipGetProcAddress;
 	fn_hipGetErrorString hipGetErrorString;
+	fn_hipMemCreate hipMemCreate;
+	fn_hipMemRelease hipMemRelease;
+	fn_hipMemMap hipMemMap;
+	fn_hipMemUnmap hipMemUnmap;
 };
 
 extern struct nvs_real_hip real;

@@ -142,6 +142,18 @@ typedef nvshipError_t (*fn_hipGetProcAddress)(const char *, void **, int,
 					      uint64_t, void *);
 typedef const char *(*fn_hipGetErrorString)(nvshipError_t);
 
+/* Virtual-memory-management API (PyTorch expandable segments).
+ * hipMemGenericAllocationHandle_t is an opaque unsigned long long;
+ * hipMemAllocationProp is opaque to us (const pointer pass-through). */
+typedef unsigned long long nvship_memhandle_t;
+typedef nvshipError_t (*fn_hipMemCreate)(nvship_memhandle_t *, size_t,
+					 const void *, unsigned long long);
+typedef nvshipError_t (*fn_hipMemRelease)(nvship_memhandle_t);
+typedef nvshipError_t (*fn_hipMemMap)(void *, size_t, size_t,
+				      nvship_memhandle_t,
+				      unsigned long long);
+typedef nvshipError_t (*fn_hipMemUnmap)(void *, size_t);
+
 /* ROCm SMI (librocm_smi64.so), for idle detection. */
 typedef int (*fn_rsmi_init)(uint64_t);
 typedef int (*fn_rsmi_dev_busy_percent_get)(uint32_t, uint32_t *);

@@ -123,6 +123,35 @@ hipError_t hipMallocFromPoolAsync(void **ptr, size_t size, void *pool,
 	return do_alloc(ptr, size);
 }
 
+/* VMM API (expandable segments): handle = the malloc'd block. */
+hipError_t hipMemCreate(unsigned long long *handle, size_t size,
+			const void *prop, unsigned long long flags)
+{
+	void *p = NULL;
+	hipError_t r;
+
+	(void)prop; (void)flags;
+	ev("hipMemCreate", (long long)size);
+	r = do_alloc(&p, size);
+	*handle = (unsigned long long)(uintptr_t)p;
+	return r;
+}
+
+hipError_t hipMemRelease(unsigned long long handle)
+{
+	ev("hipMemRelease", (long long)handle);
+	free((void *)(uintptr_t)handle);
+	return 0;
+}
+
+hipError_t hipMemMap(void *ptr, size_t size, size_t offset,
+		     unsigned long long handle, unsigned long long flags)
+{
+	(void)ptr; (void)size; (void)offset; (void)handle; (void)flags;
+	ev("hipMemMap", (long long)size);
+	return 0;
+}
+
 hipError_t hipFree(void *ptr)
 {
 	ev("hipFree", (long long)(uintptr_t)ptr);

@@ -195,6 +195,10 @@ static void bootstrap(void)
 		resolve("hipModuleLaunchCooperativeKernel");
 	real.hipGetProcAddress =
 		(fn_hipGetProcAddress)resolve("hipGetProcAddress");
+	real.hipMemCreate = (fn_hipMemCreate)resolve("hipMemCreate");
+	real.hipMemRelease = (fn_hipMemRelease)resolve("hipMemRelease");
+	real.hipMemMap = (fn_hipMemMap)resolve("hipMemMap");
+	real.hipMemUnmap = (fn_hipMemUnmap)resolve("hipMemUnmap");
 	real.hipGetErrorString =
 		(fn_hipGetErrorString)resolve("hipGetErrorString");
 
@@ -687,6 +691,7 @@ enum hook_id {
 	H_hipMemcpyToSymbol, H_hipMemcpyFromSymbol, H_hipMemcpyPeerAsync,
 	H_hipMemset, H_hipMemsetAsync, H_hipMemsetD32Async,
 	H_hipGetProcAddress, H_dlsym,
+	H_hipMemCreate, H_hipMemRelease, H_hipMemMap,
 	H_COUNT_
 };
 
@@ -704,6 +709,7 @@ static const char *hook_names[H_COUNT_] = {
 	"hipMemcpyToSymbol", "hipMemcpyFromSymbol", "hipMemcpyPeerAsync",
 	"hipMemset", "hipMemsetAsync", "hipMemsetD32Async",
 	"hipGetProcAddress", "dlsym",
+	"hipMemCreate", "hipMemRelease", "hipMemMap",
 };
 
 static unsigned long hook_counts[H_COUNT_];
@@ -1239,6 +1245,112 @@ nvshipError_t hipModuleLaunchCooperativeKernel(
 		f, gx, gy, gz, bx, by, bz, shmem, stream, params));
 }
 
+/*
+ * Virtual-memory-management path (PyTorch
+ * PYTORCH_HIP_ALLOC_CONF=expandable_segments:True allocates via
+ * hipMemCreate/hipMemMap and would otherwise bypass every invariant;
+ * the reference never faced it — CUDA 11.1-era PyTorch, SURVEY.md §7
+ * step 4).  Physical VMM handles cannot be converted to managed
+ * memory, so these allocations stay real VRAM: they are COUNTED
+ * against the cap (handle-keyed) and loudly flagged as
+ * non-oversubscribable.  client_env() additionally strips
+ * expandable_segments from the allocator config so PyTorch under the
+ * device plugin never takes this path by default.
+ */
+struct nvs_vmm_alloc {
+	nvship_memhandle_t handle;
+	size_t size;
+	struct nvs_vmm_alloc *next;
+};
+static struct nvs_vmm_alloc *vmm_list;
+static pthread_mutex_t vmm_mutex = PTHREAD_MUTEX_INITIALIZER;
+
+nvshipError_t hipMemCreate(nvship_memhandle_t *handle, size_t size,
+			   const void *prop, unsigned long long flags)
+{
+	static int warned;
+	nvshipError_t r;
+
+	BOOTSTRAP();
+	BUMP(H_hipMemCreate);
+	CHECK_REAL(hipMemCreate);
+	if (!disable_um && !warned) {
+		warned = 1;
+		log_warn("hipMemCreate (VMM / expandable segments) in use: "
+			 "these allocations are capped but stay in real "
+			 "VRAM and cannot be oversubscribed or migrated; "
+			 "prefer the default PyTorch allocator under "
+			 "nvshare");
+	}
+	if (!disable_um && !oversub_allowed && size > 0 &&
+	    reserve_cap(size) != 0)
+		return NVSHIP_ERROR_OOM;
+	r = real.hipMemCreate(handle, size, prop, flags);
+	if (r == NVSHIP_SUCCESS && !disable_um && size > 0) {
+		struct nvs_vmm_alloc *a = malloc(sizeof(*a));
+
+		if (a != NULL) {
+			a->handle = *handle;
+			a->size = size;
+			pthread_mutex_lock(&vmm_mutex);
+			a->next = vmm_list;
+			vmm_list = a;
+			pthread_mutex_unlock(&vmm_mutex);
+			if (oversub_allowed) {
+				pthread_mutex_lock(&alloc_mutex);
+				sum_allocated += size;
+				pthread_mutex_unlock(&alloc_mutex);
+			}
+			log_debug("hook: +vmm %llu %zu MiB (sum %zu MiB)",
+				  (unsigned long long)*handle,
+				  size / NVS_MIB, sum_allocated / NVS_MIB);
+		}
+	} else if (r != NVSHIP_SUCCESS && !disable_um && !oversub_allowed &&
+		   size > 0) {
+		unreserve_cap(size);
+	}
+	return r;
+}
+
+nvshipError_t hipMemRelease(nvship_memhandle_t handle)
+{
+	struct nvs_vmm_alloc **pp, *a;
+	size_t size = 0;
+
+	BOOTSTRAP();
+	BUMP(H_hipMemRelease);
+	CHECK_REAL(hipMemRelease);
+	pthread_mutex_lock(&vmm_mutex);
+	for (pp = &vmm_list; *pp != NULL; pp = &(*pp)->next) {
+		if ((*pp)->handle == handle) {
+			a = *pp;
+			*pp = a->next;
+			size = a->size;
+			free(a);
+			break;
+		}
+	}
+	pthread_mutex_unlock(&vmm_mutex);
+	if (size != 0) {
+		pthread_mutex_lock(&alloc_mutex);
+		sum_allocated -= size;
+		pthread_mutex_unlock(&alloc_mutex);
+		log_debug("hook: -vmm %llu %zu MiB (sum %zu MiB)",
+			  (unsigned long long)handle, size / NVS_MIB,
+			  sum_allocated / NVS_MIB);
+	}
+	return real.hipMemRelease(handle);
+}
+
+nvshipError_t hipMemMap(void *ptr, size_t size, size_t offset,
+			nvship_memhandle_t handle, unsigned long long flags)
+{
+	BOOTSTRAP();
+	BUMP(H_hipMemMap);
+	CHECK_REAL(hipMemMap);
+	return real.hipMemMap(ptr, size, offset, handle, flags);
+}
+
 /* Direct managed allocations by the app: tracked and capped too. */
 nvshipError_t hipMallocManaged(void **ptr, size_t size, unsigned int flags)
 {
@@ -1310,6 +1422,9 @@ static const struct hook_entry hook_table[] = {
 	{ "hipModuleLaunchCooperativeKernel",
 	  (void *)hipModuleLaunchCooperativeKernel },
 	{ "hipMallocManaged", (void *)hipMallocManaged },
+	{ "hipMemCreate", (void *)hipMemCreate },
+	{ "hipMemRelease", (void *)hipMemRelease },
+	{ "hipMemMap", (void *)hipMemMap },
 	/* dlsym-resolved hipGetProcAddress must return OUR wrapper,
 	 * else every entry point fetched through it silently bypasses
 	 * the cap and the gate (the reference hooked its

@@ -243,6 +243,59 @@ int main(void) {
     assert chunks == 12, (chunks, r.stderr[-2000:])
 
 
+def test_vmm_alloc_counted_against_cap(artifacts, sched, sock_dir):
+    """hipMemCreate (PyTorch expandable-segments backend) cannot dodge
+    the cap: physical VMM chunks reserve against the limit and
+    hipMemRelease returns the headroom."""
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     reserve_mib=16, fake_total_mib=64)
+    env["NVSTUB_TOTAL_MIB"] = "4096"
+    code = r"""
+#include <stdio.h>
+typedef int hipError_t;
+extern hipError_t hipMemCreate(unsigned long long *, unsigned long,
+                               const void *, unsigned long long);
+extern hipError_t hipMemRelease(unsigned long long);
+int main(void) {
+    unsigned long long h[100]; int n = 0;
+    while (n < 100 && hipMemCreate(&h[n], 16ul << 20, 0, 0) == 0)
+        n++;
+    printf("CHUNKS=%d\n", n);           /* limit 48 MiB -> 3 x 16 */
+    if (n > 0) hipMemRelease(h[0]);     /* room for exactly one more */
+    unsigned long long h2;
+    printf("AFTER_RELEASE=%d\n", hipMemCreate(&h2, 16ul << 20, 0, 0));
+    return 0;
+}
+"""
+    src = os.path.join(sock_dir, "vmm.c")
+    exe = os.path.join(sock_dir, "vmm")
+    with open(src, "w") as f:
+        f.write(code)
+    build = subprocess.run(
+        ["gcc", "-o", exe, src, "-L", str(artifacts.stub_dir),
+         "-lamdhip64", f"-Wl,-rpath,{artifacts.stub_dir}"],
+        capture_output=True, text=True)
+    assert build.returncode == 0, build.stderr
+    r = subprocess.run([exe], env=env, capture_output=True, text=True,
+                       timeout=60)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    assert "CHUNKS=3" in r.stdout, (r.stdout, r.stderr[-2000:])
+    assert "AFTER_RELEASE=0" in r.stdout, (r.stdout, r.stderr[-2000:])
+    assert "expandable segments" in r.stderr  # loud one-time warning
+
+
+def test_client_env_strips_expandable_segments():
+    """client_env removes the expandable-segments allocator option so
+    PyTorch under the plugin uses the (managed-convertible) default."""
+    base = {"PYTORCH_HIP_ALLOC_CONF":
+            "expandable_segments:True,max_split_size_mb:256"}
+    env = client_env(base=base)
+    assert env["PYTORCH_HIP_ALLOC_CONF"] == "max_split_size_mb:256"
+    base = {"PYTORCH_CUDA_ALLOC_CONF": "expandable_segments:True"}
+    env = client_env(base=base)
+    assert "PYTORCH_CUDA_ALLOC_CONF" not in env
+
+
 def test_fork_safety(artifacts, sched, sock_dir):
     """A forked child free-runs without corrupting the parent's
     scheduler protocol or deadlocking (reference would deadlock)."""
