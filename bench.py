@@ -183,6 +183,27 @@ def main(argv=None):
     if world > 1:
         dist.barrier()
 
+    # Solo reference phase: rank 0 alone times a few steps while the
+    # other ranks idle (their clients early-release the lock), giving
+    # the denominator for the per-job-slowdown% half of the headline
+    # metric (BASELINE.json: "makespan + per-job slowdown%").
+    solo_ms = None
+    k_solo = max(3, args.steps // 4)
+    if rank == 0:
+        if device == "cuda":
+            time.sleep(0.5)  # let idle ranks hand the lock over
+        ts = time.monotonic()
+        for _ in range(k_solo):
+            step()
+        if device == "cuda":
+            torch.cuda.synchronize()
+        solo_ms = (time.monotonic() - ts) * 1000.0 / k_solo
+    if world > 1:
+        solo_l = [None]
+        dist.broadcast_object_list(solo_l, src=0)
+        solo_ms = solo_l[0] if rank != 0 else solo_ms
+        dist.barrier()
+
     t0 = time.monotonic()
     for _ in range(args.steps):
         loss = step()
@@ -230,6 +251,16 @@ def main(argv=None):
                     + (", stock (no interposer)" if args.stock else ""),
                 "per_rank_seconds": [round(t, 3) for t in times],
                 "makespan_s": round(makespan, 3),
+                # Per-job slowdown vs the ideal 1/N time share:
+                # co-located per-rank time / (N x solo time) - 1.
+                "solo_ms_per_step": round(solo_ms, 3),
+                "per_job_slowdown_pct": [
+                    round((t * 1000.0 / args.steps)
+                          / (world * solo_ms) * 100.0 - 100.0, 1)
+                    for t in times],
+                "sharing_efficiency": round(
+                    world * solo_ms
+                    / (tmax * 1000.0 / args.steps), 4),
                 "loss": round(final_loss, 4),
                 "oversub_fake_mib": args.oversub_fake_mib,
             },
