@@ -58,34 +58,75 @@ GATED = [
 ]
 
 
+def _csv_counts(path: str) -> dict[str, int]:
+    """Extract hip* API counts from one rocprofv3 CSV, whatever its
+    exact flavor: a stats file (NAME + CALLS columns) yields the call
+    totals; a per-event trace (a name/function column, no calls)
+    yields one count per row."""
+    counts: dict[str, int] = {}
+    with open(path, newline="") as f:
+        reader = csv.reader(f)
+        try:
+            header = next(reader)
+        except StopIteration:
+            return counts
+        cols = {c.strip().strip('"').lower(): i
+                for i, c in enumerate(header)}
+        name_i = next((cols[k] for k in ("name", "function", "kind")
+                       if k in cols), None)
+        calls_i = next((cols[k] for k in ("calls", "count")
+                        if k in cols), None)
+        if name_i is None:
+            return counts
+        for row in reader:
+            if name_i >= len(row):
+                continue
+            name = row[name_i].strip().strip('"')
+            if not name.startswith("hip"):
+                continue
+            n = 1
+            if calls_i is not None and calls_i < len(row):
+                try:
+                    n = int(float(row[calls_i]))
+                except ValueError:
+                    continue
+            counts[name] = counts.get(name, 0) + n
+    return counts
+
+
 def rocprof_counts(workdir: str) -> dict[str, int]:
     env = dict(os.environ)
     env["HSA_XNACK"] = "1"
     r = subprocess.run(
         ["rocprofv3", "--runtime-trace", "--stats", "-d", workdir,
+         "--output-format", "csv",
          "--", sys.executable, "-c", WORKLOAD],
         capture_output=True, text=True, timeout=900, env=env,
         cwd=workdir)
     assert "WORKLOAD_OK" in r.stdout, (r.stdout[-2000:],
                                        r.stderr[-2000:])
+    all_csvs = glob.glob(os.path.join(workdir, "**", "*.csv"),
+                         recursive=True)
+    # Prefer stats CSVs (exact totals); fall back to counting trace
+    # rows; a domain-stats table printed to stdout is last resort.
+    stats = [p for p in all_csvs if "stats" in os.path.basename(p)
+             and "api" in os.path.basename(p)]
+    traces = [p for p in all_csvs if "api" in os.path.basename(p)
+              and p not in stats]
     counts: dict[str, int] = {}
-    # rocprofv3 writes *_hip_api_stats.csv ("Name","Calls",...) and/or
-    # prints a stats table; prefer the CSV.
-    for path in glob.glob(os.path.join(workdir, "**", "*hip_api*.csv"),
-                          recursive=True):
-        with open(path) as f:
-            for row in csv.DictReader(f):
-                name = (row.get("Name") or row.get("NAME") or "")
-                calls = row.get("Calls") or row.get("CALLS") or "0"
-                name = name.strip().strip('"')
-                if name:
-                    counts[name] = counts.get(name, 0) + int(calls)
+    for path in stats or traces:
+        for k, v in _csv_counts(path).items():
+            counts[k] = counts.get(k, 0) + v
     if not counts:
-        # fallback: parse the printed stats table
         for line in (r.stdout + r.stderr).splitlines():
             m = re.match(r'\s*"?(hip\w+)"?\s*[|,]\s*(\d+)', line)
             if m:
                 counts[m.group(1)] = int(m.group(2))
+    # A 0-vs-0 comparison proves nothing (round-1 overclaim): the
+    # crosscheck is only valid when the profiler actually saw APIs.
+    assert counts, ("rocprofv3 produced no parsable API counts; "
+                    f"files: {all_csvs}; stdout tail: "
+                    f"{r.stdout[-1500:]}; stderr tail: {r.stderr[-1500:]}")
     return counts
 
 
@@ -126,12 +167,14 @@ def main() -> None:
         if p > 0 and hk == 0:
             mismatches.append(api)
     res = {"rows": rows, "uncovered_apis": mismatches,
-           "rocprof_total_apis": len(prof)}
+           "rocprof_total_apis": len(prof),
+           "rocprof_total_calls": sum(prof.values())}
     out = Path(args.out)
     out.parent.mkdir(parents=True, exist_ok=True)
     out.write_text(json.dumps(res, indent=2))
-    print(json.dumps({"uncovered": mismatches}, indent=2))
-    if mismatches:
+    print(json.dumps({"uncovered": mismatches,
+                      "rocprof_total_apis": len(prof)}, indent=2))
+    if mismatches or not prof:
         sys.exit(1)
 
 
