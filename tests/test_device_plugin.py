@@ -195,6 +195,60 @@ def test_restart_on_kubelet_socket_recreation(kubelet_dir):
         p.stop()
 
 
+def test_reregister_after_kubelet_outage(kubelet_dir):
+    """Socket deleted with no kubelet for a while: the plugin keeps
+    retrying and re-registers when the kubelet comes back (reference
+    main.go restart loop survives kubelet downtime)."""
+    registrations = []
+    ev = threading.Event()
+
+    def register(req_bytes, ctx):
+        registrations.append(pb.RegisterRequest.FromString(req_bytes))
+        ev.set()
+        return pb.Empty()
+
+    handler = grpc.method_handlers_generic_handler(
+        pb.REGISTRATION_SERVICE,
+        {"Register": grpc.unary_unary_rpc_method_handler(
+            register,
+            response_serializer=lambda m: m.SerializeToString())})
+
+    def make_kubelet():
+        server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+        server.add_generic_rpc_handlers((handler,))
+        server.add_insecure_port(f"unix://{kubelet_dir}/kubelet.sock")
+        server.start()
+        return server
+
+    k1 = make_kubelet()
+    cfg = PluginConfig(virtual_devices=2, kubelet_dir=kubelet_dir,
+                       gpus=[0])
+    p = DevicePlugin(cfg)
+    t = threading.Thread(target=p.run_forever, daemon=True)
+    t.start()
+    try:
+        assert ev.wait(10), "first registration missing"
+        ev.clear()
+        n_first = len(registrations)
+        time.sleep(1.0)
+        # Kubelet dies and its socket disappears; nothing replaces it
+        # for several seconds.
+        k1.stop(grace=0)
+        import contextlib
+        import os as _os
+        with contextlib.suppress(FileNotFoundError):
+            _os.unlink(f"{kubelet_dir}/kubelet.sock")
+        time.sleep(5.0)
+        assert not ev.is_set(), "spurious registration with no kubelet"
+        # Kubelet returns: the plugin must re-register on its own.
+        k2 = make_kubelet()
+        assert ev.wait(20), "no re-registration after kubelet outage"
+        assert len(registrations) > n_first
+        k2.stop(grace=0)
+    finally:
+        p.stop()
+
+
 def test_preferred_allocation(plugin):
     req = pb.PreferredAllocationRequest()
     creq = req.container_requests.add()
