@@ -27,7 +27,7 @@ def build(model_name: str, num_classes: int):
 def run_training(model_name: str = "resnet50", device: str = "cuda",
                  batch: int = 64, image: int = 224, steps: int = 50,
                  warmup: int = 5, dtype: str = "bfloat16",
-                 num_classes: int = 1000) -> dict:
+                 num_classes: int = 1000, lr: float = 0.02) -> dict:
     import torch
 
     dev = torch.device(device)
@@ -40,7 +40,10 @@ def run_training(model_name: str = "resnet50", device: str = "cuda",
             dev, memory_format=torch.channels_last)
     else:
         model = build(model_name, num_classes).to(dev)
-    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    # lr 0.02 descends cleanly for 40+ bf16 steps on random labels
+    # (profiles/nanhunt.log stock arm: 7.17 -> 0.003); 0.1 risks
+    # early-step blowup on some seeds.
+    opt = torch.optim.SGD(model.parameters(), lr=lr, momentum=0.9)
     lossf = torch.nn.CrossEntropyLoss()
     x = torch.randn(batch, 3, image, image, device=dev)
     if device.startswith("cuda"):
@@ -86,10 +89,11 @@ def main(argv: list[str] | None = None) -> None:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--dtype", default="bfloat16")
     ap.add_argument("--num-classes", type=int, default=1000)
+    ap.add_argument("--lr", type=float, default=0.02)
     args = ap.parse_args(argv)
     res = run_training(args.model, args.device, args.batch, args.image,
                        args.steps, args.warmup, args.dtype,
-                       args.num_classes)
+                       args.num_classes, args.lr)
     res["label"] = args.label
     emit(res)
 

@@ -96,8 +96,11 @@ static size_t mem_total;        /* advertised total (bytes) */
 static size_t mem_reserve;      /* carve-out (bytes) */
 static int oversub_allowed;
 static int disable_um;
-static int alloc_prefetch = 1;   /* NVSHARE_ALLOC_PREFETCH */
+static int alloc_prefetch = 0;   /* NVSHARE_ALLOC_PREFETCH (see
+				  * populate_managed: corrupts on ROCm
+				  * 7.2, default off) */
 static int coarse_grain = 1;     /* NVSHARE_COARSE_GRAIN */
+static int preferred_loc = 1;    /* NVSHARE_PREFERRED_LOC */
 static pthread_once_t memquery_once = PTHREAD_ONCE_INIT;
 
 /* pending-kernel window */
@@ -217,8 +220,9 @@ static void bootstrap(void)
 
 	oversub_allowed = nvs_env_bool("NVSHARE_ENABLE_SINGLE_OVERSUB", 0);
 	disable_um = nvs_env_bool("NVSHARE_DISABLE_UM", 0);
-	alloc_prefetch = nvs_env_bool("NVSHARE_ALLOC_PREFETCH", 1);
+	alloc_prefetch = nvs_env_bool("NVSHARE_ALLOC_PREFETCH", 0);
 	coarse_grain = nvs_env_bool("NVSHARE_COARSE_GRAIN", 1);
+	preferred_loc = nvs_env_bool("NVSHARE_PREFERRED_LOC", 1);
 	/* Reserve sized for 288 GB HBM3E; the reference used 1536 MiB on a
 	 * 16 GB P100 (hook.c:45). */
 	mem_reserve = (size_t)nvs_env_long("NVSHARE_RESERVE_MIB", 8192, 0,
@@ -286,31 +290,36 @@ static size_t mem_limit(void)
 #define NVSHIP_MEM_ADVISE_SET_COARSE_GRAIN 100
 
 /*
- * gfx950 managed-memory fast path.  Measured on MI355X
- * (tools/faultbench.py, profiles/faultbench.json): naive fine-grain
- * XNACK device-first-touch runs at ~0.09 GB/s — 4 GB of fresh managed
- * memory costs ~44 s of page faults — while prefetch-populated memory
- * touches at HBM speed.  So every converted allocation is (a) advised
- * coarse-grain (whole-range migration granularity, full-rate access;
- * fine-grain host/device *concurrent* access is not something the
- * hipMalloc contract we replace ever promised) and (b) prefetched to
- * the device so pages are born resident instead of demand-faulted one
- * 4 KiB page at a time.
+ * gfx950 managed-memory fast path.
  *
- * The prefetch is GPU (SDMA) work and therefore gated: a co-located
- * client populating 6 GB mid-quantum forcibly evicts the lock
- * holder's resident pages and the two migration flows livelock at
- * fault speed (observed: a 26 s experiment became a >1450 s hang).
- * If we hold the lock at hipMalloc time, populate immediately;
- * otherwise the range is marked pending and populated at the first
- * gated submission (nvs_populate_pending, called under the lock).
- * Returns 1 when populated now.
+ * Every converted allocation is advised (a) coarse-grain (whole-range
+ * migration granularity, full-rate access, working hardware FP
+ * atomics; fine-grain host/device *concurrent* access is not
+ * something the hipMalloc contract we replace ever promised) and (b)
+ * preferred-location = the app's device, so first-touch pages are
+ * born in HBM instead of migrating later.  Both advise calls are
+ * metadata-only: no data moves, nothing can race the app's writes.
+ *
+ * An earlier design ALSO issued an eager hipMemPrefetchAsync here
+ * (the round-1 NVSHARE_ALLOC_PREFETCH=1 default).  Measured on ROCm
+ * 7.2 / MI355X that CORRUPTS the allocation: the null-stream prefetch
+ * races PyTorch's initialization writes on non-blocking streams and
+ * loses some of them — ResNet-50 weights came up garbage (initial
+ * loss 930 vs 7.17) or trained to NaN within 3 steps, with bad grads
+ * in every layer (profiles/nanhunt.log, tools/nanhunt.{py,sh}: every
+ * arm with alloc-prefetch on corrupts, every arm without it matches
+ * stock exactly).  The env survives for experiments but defaults OFF.
+ * Returns 1 when populated now (0 = deferred to the gate).
  */
 static int populate_managed(void *ptr, size_t size)
 {
 	if (coarse_grain && real.hipMemAdvise != NULL)
 		real.hipMemAdvise(ptr, size,
 				  NVSHIP_MEM_ADVISE_SET_COARSE_GRAIN, 0);
+	if (preferred_loc && real.hipMemAdvise != NULL)
+		real.hipMemAdvise(ptr, size,
+				  NVSHIP_MEM_ADVISE_SET_PREFERRED_LOCATION,
+				  nvs_app_device);
 	if (!alloc_prefetch || real.hipMemPrefetchAsync == NULL)
 		return 1; /* nothing to defer */
 	if (!nvs_can_submit_now())
