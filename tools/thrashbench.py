@@ -15,10 +15,13 @@ Usage (GPU box):
 from __future__ import annotations
 
 import argparse
+import glob as _glob
 import json
 import subprocess
 import sys
 import tempfile
+import threading
+import time
 from pathlib import Path
 
 REPO = Path(__file__).resolve().parent.parent
@@ -27,6 +30,48 @@ sys.path.insert(0, str(REPO))
 from nvshare_amd import ctl  # noqa: E402
 from nvshare_amd.colocate import run_colocated, workload_cmd  # noqa: E402
 from nvshare_amd.scheduler import SchedulerDaemon  # noqa: E402
+
+
+class VramSampler:
+    """Peak VRAM (MiB) over a window, from sysfs (thesis Tables
+    11.7/11.8 report peak GPU memory per scenario)."""
+
+    def __init__(self, period_s: float = 0.5):
+        self.paths = _glob.glob(
+            "/sys/class/drm/card*/device/mem_info_vram_used")
+        self.peak = 0
+        self.period = period_s
+        self._stop = threading.Event()
+        self._t = threading.Thread(target=self._run, daemon=True)
+
+    def _read(self) -> int:
+        total = 0
+        for p in self.paths:
+            try:
+                with open(p) as f:
+                    total += int(f.read().strip())
+            except OSError:
+                pass
+        return total
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            v = self._read()
+            if v > self.peak:
+                self.peak = v
+            self._stop.wait(self.period)
+
+    def __enter__(self):
+        self._t.start()
+        return self
+
+    def __exit__(self, *exc):
+        self._stop.set()
+        self._t.join(timeout=2)
+
+    @property
+    def peak_mib(self) -> int:
+        return self.peak // (1024 * 1024)
 
 
 def main() -> None:
@@ -68,24 +113,30 @@ def main() -> None:
                 str(batch), "--image", str(args.image), "--steps",
                 str(args.steps), "--warmup", "5")
             row: dict = {"batch": batch}
-            solo = run_colocated([cmd], sock_dir=sock_dir,
-                                 env_kwargs=env_kwargs,
-                                 timeout=args.timeout)
+            with VramSampler() as vs:
+                solo = run_colocated([cmd], sock_dir=sock_dir,
+                                     env_kwargs=env_kwargs,
+                                     timeout=args.timeout)
             row["solo_s"] = solo.makespan if solo.ok else None
-            two = run_colocated([cmd, cmd], sock_dir=sock_dir,
-                                env_kwargs=env_kwargs,
-                                timeout=args.timeout)
+            row["solo_peak_mib"] = vs.peak_mib
+            with VramSampler() as vs:
+                two = run_colocated([cmd, cmd], sock_dir=sock_dir,
+                                    env_kwargs=env_kwargs,
+                                    timeout=args.timeout)
             row["parallel2_s"] = two.makespan if two.ok else None
+            row["parallel2_peak_mib"] = vs.peak_mib
             if row["solo_s"] and row["parallel2_s"]:
                 row["parallel_vs_serial"] = (
                     row["parallel2_s"] / (2 * row["solo_s"]))
             if args.include_off:
                 ctl.set_scheduling(False, sock_dir)
-                off = run_colocated([cmd, cmd], sock_dir=sock_dir,
-                                    env_kwargs=env_kwargs,
-                                    timeout=args.off_timeout)
+                with VramSampler() as vs:
+                    off = run_colocated([cmd, cmd], sock_dir=sock_dir,
+                                        env_kwargs=env_kwargs,
+                                        timeout=args.off_timeout)
                 row["parallel2_sched_off_s"] = (
                     off.makespan if off.ok else None)
+                row["parallel2_sched_off_peak_mib"] = vs.peak_mib
                 ctl.set_scheduling(True, sock_dir)
             out["rows"].append(row)
             print(json.dumps(row), flush=True)
