@@ -182,8 +182,11 @@ hipError_t hipMemGetInfo(size_t *freep, size_t *totalp)
 hipError_t hipMemPrefetchAsync(const void *p, size_t n, int dev,
 			       hipStream_t s)
 {
-	(void)p; (void)dev; (void)s;
-	ev("hipMemPrefetchAsync", (long long)n);
+	(void)p; (void)s;
+	/* Direction-tagged so tests can tell eviction (to CPU, dev -1)
+	 * from restore (to device). */
+	ev(dev < 0 ? "hipMemPrefetchAsync_cpu" : "hipMemPrefetchAsync",
+	   (long long)n);
 	return 0;
 }
 

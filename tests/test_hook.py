@@ -243,6 +243,42 @@ int main(void) {
     assert chunks == 12, (chunks, r.stderr[-2000:])
 
 
+def test_pressure_verdict_drives_migration(artifacts, sched, sock_dir):
+    """Two co-located clients whose combined tracked sets exceed the
+    reported capacity get p=1 with their grants and flip migration
+    assist on WITHOUT NVSHARE_EVICT/NVSHARE_PREFETCH being set
+    (docs/roadmap.md round-1 item #2): evictions to host show up in
+    the stub event log at lock handoffs."""
+    import threading
+
+    results = []
+
+    def one(idx):
+        log = os.path.join(sock_dir, f"press{idx}.log")
+        r = run_hipclient(
+            artifacts, sock_dir, "--allocs", 4, "--alloc-mib", 100,
+            "--iters", 600, "--sleep-us", 10000, "--sync-every", 50,
+            stub_env={"NVSTUB_LOG": log, "NVSTUB_TOTAL_MIB": "600"},
+            env_extra={"NVSHARE_RELEASE_INTERVAL_MS": "200"},
+            reserve_mib=64, timeout=120)
+        results.append((r, log))
+
+    ts = [threading.Thread(target=one, args=(i,)) for i in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+
+    evicts = 0
+    for r, log in results:
+        assert r.returncode == 0, (r.stdout, r.stderr[-2000:])
+        names = [e[2] for e in load_events(log)]
+        evicts += names.count("hipMemPrefetchAsync_cpu")
+    # 2x400 MiB on a 600 MiB device: at least one handoff must have
+    # evicted the releasing client's set to host.
+    assert evicts > 0, [r.stderr[-1500:] for r, _ in results]
+
+
 def test_vmm_alloc_counted_against_cap(artifacts, sched, sock_dir):
     """hipMemCreate (PyTorch expandable-segments backend) cannot dodge
     the cap: physical VMM chunks reserve against the limit and
