@@ -39,6 +39,7 @@
 #include <stdio.h>
 #include <stdlib.h>
 #include <string.h>
+#include <sys/mman.h>
 
 #include "common.h"
 #include "hip_defs.h"
@@ -101,6 +102,11 @@ static int alloc_prefetch = 0;   /* NVSHARE_ALLOC_PREFETCH (see
 				  * 7.2, default off) */
 static int coarse_grain = 1;     /* NVSHARE_COARSE_GRAIN */
 static int preferred_loc = 1;    /* NVSHARE_PREFERRED_LOC */
+static int thp_advise = 0;       /* NVSHARE_THP: MADV_HUGEPAGE on
+				  * managed ranges (experiment knob:
+				  * 2 MiB mappings cut device TLB
+				  * pressure if the driver honors
+				  * them) */
 static pthread_once_t memquery_once = PTHREAD_ONCE_INIT;
 
 /* pending-kernel window */
@@ -223,6 +229,7 @@ static void bootstrap(void)
 	alloc_prefetch = nvs_env_bool("NVSHARE_ALLOC_PREFETCH", 0);
 	coarse_grain = nvs_env_bool("NVSHARE_COARSE_GRAIN", 1);
 	preferred_loc = nvs_env_bool("NVSHARE_PREFERRED_LOC", 1);
+	thp_advise = nvs_env_bool("NVSHARE_THP", 0);
 	/* Reserve sized for 288 GB HBM3E; the reference used 1536 MiB on a
 	 * 16 GB P100 (hook.c:45). */
 	mem_reserve = (size_t)nvs_env_long("NVSHARE_RESERVE_MIB", 8192, 0,
@@ -313,6 +320,8 @@ static size_t mem_limit(void)
  */
 static int populate_managed(void *ptr, size_t size)
 {
+	if (thp_advise)
+		madvise(ptr, size, MADV_HUGEPAGE);
 	if (coarse_grain && real.hipMemAdvise != NULL)
 		real.hipMemAdvise(ptr, size,
 				  NVSHIP_MEM_ADVISE_SET_COARSE_GRAIN, 0);
@@ -984,11 +993,72 @@ nvshipError_t hipSetDevice(int dev)
 
 /* ---- gated work submissions ---- */
 
+/*
+ * Destination prefetch for bulk writes into managed ranges.  An HtoD
+ * copy (or large memset) into FRESH managed memory otherwise runs at
+ * page-fault speed (~0.1 GB/s: measured ~17 ms per model-weight copy,
+ * gpurun_out/ab.log hk_prof arm — ~10 s of a 25 s ResNet job was
+ * memcpy wrappers at init).  Prefetching the destination range to the
+ * device ON THE SAME STREAM materializes the pages in HBM first, so
+ * the DMA lands at full rate.  Unlike the removed alloc-time eager
+ * prefetch this cannot race anything: it is ordered before the copy
+ * on its stream, and the copy's contract already forbids concurrent
+ * writers to dst.  NVSHARE_COPY_PREFETCH=0 disables.
+ */
+static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
+{
+	static int enabled = -1;
+	static long min_mib = 1;
+	struct nvs_alloc *a;
+	char *base = NULL, *d = dst;
+	size_t size = 0;
+
+	if (enabled < 0) {
+		enabled = nvs_env_bool("NVSHARE_COPY_PREFETCH", 1);
+		min_mib = nvs_env_long("NVSHARE_COPY_PREFETCH_MIB", 1, 0,
+				       1024 * 1024);
+	}
+	if (!enabled || disable_um || real.hipMemPrefetchAsync == NULL)
+		return;
+	if (n < (size_t)min_mib * NVS_MIB)
+		return;
+	pthread_mutex_lock(&alloc_mutex);
+	for (a = alloc_list; a != NULL; a = a->next) {
+		char *p = a->ptr;
+
+		if (!a->passthrough && d >= p && d < p + a->size) {
+			base = p;
+			size = a->size;
+			break;
+		}
+	}
+	pthread_mutex_unlock(&alloc_mutex);
+	if (base == NULL)
+		return;
+	if (d + n > base + size)
+		n = (size_t)(base + size - d); /* clamp to the range */
+	real.hipMemPrefetchAsync(d, n, nvs_app_device, s);
+}
+
 #define GATED2(id, call)                                                   \
 	do {                                                               \
 		nvshipError_t r_;                                          \
 		int64_t t0_ = prof_begin();                                \
 		nvs_submit_begin();                                        \
+		r_ = (call);                                               \
+		nvs_submit_end();                                          \
+		prof_end(id, t0_);                                         \
+		return r_;                                                 \
+	} while (0)
+
+/* Gated submission that first prefetches the written range (dst) to
+ * the device on the same stream. */
+#define GATED_COPY(id, dst, n, stream, call)                               \
+	do {                                                               \
+		nvshipError_t r_;                                          \
+		int64_t t0_ = prof_begin();                                \
+		nvs_submit_begin();                                        \
+		copy_prefetch((void *)(dst), (n), (stream));               \
 		r_ = (call);                                               \
 		nvs_submit_end();                                          \
 		prof_end(id, t0_);                                         \
@@ -1090,7 +1160,8 @@ nvshipError_t hipMemcpy(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpy);
 	CHECK_REAL(hipMemcpy);
-	GATED2(H_hipMemcpy, real.hipMemcpy(dst, src, n, kind));
+	GATED_COPY(H_hipMemcpy, dst, n, NULL,
+		   real.hipMemcpy(dst, src, n, kind));
 }
 
 nvshipError_t hipMemcpyAsync(void *dst, const void *src, size_t n,
@@ -1099,7 +1170,8 @@ nvshipError_t hipMemcpyAsync(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyAsync);
 	CHECK_REAL(hipMemcpyAsync);
-	GATED2(H_hipMemcpyAsync, real.hipMemcpyAsync(dst, src, n, kind, s));
+	GATED_COPY(H_hipMemcpyAsync, dst, n, s,
+		   real.hipMemcpyAsync(dst, src, n, kind, s));
 }
 
 nvshipError_t hipMemcpyWithStream(void *dst, const void *src, size_t n,
@@ -1109,7 +1181,8 @@ nvshipError_t hipMemcpyWithStream(void *dst, const void *src, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyWithStream);
 	CHECK_REAL(hipMemcpyWithStream);
-	GATED2(H_hipMemcpyWithStream, real.hipMemcpyWithStream(dst, src, n, kind, s));
+	GATED_COPY(H_hipMemcpyWithStream, dst, n, s,
+		   real.hipMemcpyWithStream(dst, src, n, kind, s));
 }
 
 nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
@@ -1118,7 +1191,8 @@ nvshipError_t hipMemcpyHtoD(nvship_deviceptr_t dst, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyHtoD);
 	CHECK_REAL(hipMemcpyHtoD);
-	GATED2(H_hipMemcpyHtoD, real.hipMemcpyHtoD(dst, src, n));
+	GATED_COPY(H_hipMemcpyHtoD, dst, n, NULL,
+		   real.hipMemcpyHtoD(dst, src, n));
 }
 
 nvshipError_t hipMemcpyDtoH(void *dst, nvship_deviceptr_t src, size_t n)
@@ -1144,7 +1218,8 @@ nvshipError_t hipMemcpyHtoDAsync(nvship_deviceptr_t dst, const void *src,
 	BOOTSTRAP();
 	BUMP(H_hipMemcpyHtoDAsync);
 	CHECK_REAL(hipMemcpyHtoDAsync);
-	GATED2(H_hipMemcpyHtoDAsync, real.hipMemcpyHtoDAsync(dst, src, n, s));
+	GATED_COPY(H_hipMemcpyHtoDAsync, dst, n, s,
+		   real.hipMemcpyHtoDAsync(dst, src, n, s));
 }
 
 nvshipError_t hipMemcpyDtoHAsync(void *dst, nvship_deviceptr_t src,
@@ -1171,7 +1246,8 @@ nvshipError_t hipMemset(void *dst, int value, size_t n)
 	BOOTSTRAP();
 	BUMP(H_hipMemset);
 	CHECK_REAL(hipMemset);
-	GATED2(H_hipMemset, real.hipMemset(dst, value, n));
+	GATED_COPY(H_hipMemset, dst, n, NULL,
+		   real.hipMemset(dst, value, n));
 }
 
 nvshipError_t hipMemsetAsync(void *dst, int value, size_t n,
@@ -1180,7 +1256,8 @@ nvshipError_t hipMemsetAsync(void *dst, int value, size_t n,
 	BOOTSTRAP();
 	BUMP(H_hipMemsetAsync);
 	CHECK_REAL(hipMemsetAsync);
-	GATED2(H_hipMemsetAsync, real.hipMemsetAsync(dst, value, n, s));
+	GATED_COPY(H_hipMemsetAsync, dst, n, s,
+		   real.hipMemsetAsync(dst, value, n, s));
 }
 
 nvshipError_t hipMemsetD32Async(nvship_deviceptr_t dst, int value,
@@ -1189,7 +1266,8 @@ nvshipError_t hipMemsetD32Async(nvship_deviceptr_t dst, int value,
 	BOOTSTRAP();
 	BUMP(H_hipMemsetD32Async);
 	CHECK_REAL(hipMemsetD32Async);
-	GATED2(H_hipMemsetD32Async, real.hipMemsetD32Async(dst, value, count, s));
+	GATED_COPY(H_hipMemsetD32Async, dst, count * 4, s,
+		   real.hipMemsetD32Async(dst, value, count, s));
 }
 
 nvshipError_t hipMemcpy2D(void *dst, size_t dpitch, const void *src,
