@@ -61,21 +61,30 @@ def run_training(model_name: str = "resnet50", device: str = "cuda",
         opt.step()
         return loss
 
-    for _ in range(warmup):
-        step()
+    # First loss before any update: a healthy random-init CE loss is
+    # ~ln(num_classes); the round-1 alloc-prefetch corruption showed up
+    # here as ~930 (profiles/nanhunt.log).  Recorded so callers can
+    # assert sanity, not just finiteness.
+    first_loss = None
+    for i in range(warmup):
+        loss = step()
+        if i == 0:
+            first_loss = float(loss.detach().float().cpu())
     sync(device)
     with Timer() as t:
         for _ in range(steps):
             loss = step()
         sync(device)
     final_loss = float(loss.detach().float().cpu())
+    if first_loss is None:
+        first_loss = final_loss
     return {
         "workload": "train_resnet", "model": model_name,
         "seconds": t.seconds, "steps": steps, "batch": batch,
         "image": image, "device": device, "dtype": dtype,
         "steps_per_s": steps / t.seconds,
         "samples_per_s": steps * batch / t.seconds,
-        "loss": final_loss,
+        "loss": final_loss, "loss_first": first_loss,
     }
 
 
