@@ -84,6 +84,8 @@ struct nvs_alloc {
 	void *ptr;
 	size_t size;
 	int populated;   /* device pages materialized (eager prefetch) */
+	int bulk_ready;  /* whole range residency established by the
+			  * first-bulk-write prefetch (copy_prefetch) */
 	int passthrough; /* real-VRAM (stream-ordered) alloc: counted
 			  * against the cap but never migrated */
 	struct nvs_alloc *next;
@@ -351,6 +353,7 @@ static void track_alloc2(void *ptr, size_t size, int populated,
 	a->ptr = ptr;
 	a->size = size;
 	a->populated = populated;
+	a->bulk_ready = 0;
 	a->passthrough = passthrough;
 	pthread_mutex_lock(&alloc_mutex);
 	a->next = alloc_list;
@@ -994,31 +997,40 @@ nvshipError_t hipSetDevice(int dev)
 /* ---- gated work submissions ---- */
 
 /*
- * Destination prefetch for bulk writes into managed ranges.  An HtoD
- * copy (or large memset) into FRESH managed memory otherwise runs at
- * page-fault speed (~0.1 GB/s: measured ~17 ms per model-weight copy,
- * gpurun_out/ab.log hk_prof arm — ~10 s of a 25 s ResNet job was
- * memcpy wrappers at init).  Prefetching the destination range to the
- * device ON THE SAME STREAM materializes the pages in HBM first, so
- * the DMA lands at full rate.  Unlike the removed alloc-time eager
- * prefetch this cannot race anything: it is ordered before the copy
- * on its stream, and the copy's contract already forbids concurrent
- * writers to dst.  NVSHARE_COPY_PREFETCH=0 disables.
+ * First-bulk-write residency for managed ranges.  An HtoD copy (or
+ * large memset) into FRESH managed memory runs at page-fault speed
+ * (~0.1 GB/s: ~17 ms per model-weight copy; ~10 s of a 25 s ResNet
+ * job was memcpy wrappers at init — profiles/ab_r2.log hk_prof arm).
+ *
+ * SAFETY (measured, ROCm 7.2 gfx950): hipMemPrefetchAsync's
+ * migration is NOT ordered with subsequent work enqueued on the same
+ * stream — a prefetch issued right before the copy corrupts the data
+ * just like the alloc-time variant did (gpurun_out/ab2.log: every
+ * arm with the naive same-stream prefetch trained to NaN at 10x
+ * slowdown).  The only safe pattern is prefetch + HOST SYNC before
+ * any dependent write — exactly what the LOCK_OK restore path does.
+ * So: on the FIRST large write into a tracked range, prefetch the
+ * WHOLE range, hipStreamSynchronize, and mark it; later writes into
+ * that range take the flag fast path.  One sync per allocation
+ * amortizes over the many per-tensor uploads a model load does into
+ * the same caching-allocator segment.  NVSHARE_COPY_PREFETCH=0
+ * disables.
  */
 static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 {
 	static int enabled = -1;
-	static long min_mib = 1;
+	static long min_mib = 8;
 	struct nvs_alloc *a;
 	char *base = NULL, *d = dst;
 	size_t size = 0;
 
 	if (enabled < 0) {
 		enabled = nvs_env_bool("NVSHARE_COPY_PREFETCH", 1);
-		min_mib = nvs_env_long("NVSHARE_COPY_PREFETCH_MIB", 1, 0,
+		min_mib = nvs_env_long("NVSHARE_COPY_PREFETCH_MIB", 8, 1,
 				       1024 * 1024);
 	}
-	if (!enabled || disable_um || real.hipMemPrefetchAsync == NULL)
+	if (!enabled || disable_um || real.hipMemPrefetchAsync == NULL ||
+	    real.hipStreamSynchronize == NULL)
 		return;
 	if (n < (size_t)min_mib * NVS_MIB)
 		return;
@@ -1027,17 +1039,26 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 		char *p = a->ptr;
 
 		if (!a->passthrough && d >= p && d < p + a->size) {
-			base = p;
-			size = a->size;
+			if (!a->bulk_ready) {
+				a->bulk_ready = 1;
+				base = p;
+				size = a->size;
+			}
 			break;
 		}
 	}
 	pthread_mutex_unlock(&alloc_mutex);
 	if (base == NULL)
 		return;
-	if (d + n > base + size)
-		n = (size_t)(base + size - d); /* clamp to the range */
-	real.hipMemPrefetchAsync(d, n, nvs_app_device, s);
+	/* Quiesce the device first: an in-flight write into the same
+	 * range from ANOTHER stream would race the migration (the
+	 * documented-unsafe overlap).  Once per range, so the full
+	 * drain amortizes. */
+	if (real.hipDeviceSynchronize != NULL)
+		real.hipDeviceSynchronize();
+	real.hipMemPrefetchAsync(base, size, nvs_app_device, s);
+	real.hipStreamSynchronize(s); /* migration MUST complete before
+				       * the write lands (see above) */
 }
 
 #define GATED2(id, call)                                                   \
