@@ -1025,7 +1025,14 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 	size_t size = 0;
 
 	if (enabled < 0) {
-		enabled = nvs_env_bool("NVSHARE_COPY_PREFETCH", 1);
+		/* Default OFF: even the quiesce+prefetch+host-sync
+		 * pattern corrupts when the range carries coarse-grain
+		 * or preferred-location advise (gpurun_out/ab3.log:
+		 * hk_plain clean, hooked/hk_nocg NaN) — on ROCm 7.2
+		 * hipMemPrefetchAsync of never-touched advised ranges
+		 * is not safe in-band at all.  Kept as an experiment
+		 * knob for future driver versions. */
+		enabled = nvs_env_bool("NVSHARE_COPY_PREFETCH", 0);
 		min_mib = nvs_env_long("NVSHARE_COPY_PREFETCH_MIB", 8, 1,
 				       1024 * 1024);
 	}
