@@ -1,0 +1,48 @@
+#!/bin/bash
+# Attribution round 5: is the steady-state tail (100-215 ms fault-storm
+# kernel calls) caused by per-step >256MiB pool-workspace conversions?
+set -x
+mkdir -p gpurun_out
+export HSA_XNACK=1 MIOPEN_FIND_MODE=FAST
+
+TRAIN='
+import sys; sys.path.insert(0, ".")
+from nvshare_amd.workloads.train_resnet import run_training
+import math
+r = run_training("resnet50", "cuda", batch=32, image=224, steps=60,
+                 warmup=10)
+assert math.isfinite(r["loss"]), r
+assert r["loss_first"] < 20, r
+print("ARM", round(r["samples_per_s"], 1), "loss", round(r["loss"], 3))
+'
+timeout 300 env -u LD_PRELOAD python -c "
+import sys; sys.path.insert(0, '.')
+from nvshare_amd.workloads.train_resnet import run_training
+run_training('resnet50', 'cuda', batch=32, image=224, steps=3, warmup=3)
+print('WARMED')" > gpurun_out/ab5_warm.log 2>&1
+
+: > gpurun_out/ab5.log
+run_arm () {
+  local label="$1" pre="$2" envs="$3"
+  timeout 200 env $envs $pre python -c "$TRAIN" 2>&1 \
+    | sed "s/^ARM/$label/" >> gpurun_out/ab5.log
+  echo "$label rc=${PIPESTATUS[0]}" >> gpurun_out/ab5.log
+}
+STOCK="env -u LD_PRELOAD"
+HOOK="python -m nvshare_amd.run --standalone --"
+
+run_arm stock1      "$STOCK" "IGNORE="
+run_arm hooked1     "$HOOK"  "IGNORE="
+run_arm hk_ptall    "$HOOK"  "NVSHARE_PASSTHROUGH_MIB=1048576"
+run_arm hk_ptall2   "$HOOK"  "NVSHARE_PASSTHROUGH_MIB=1048576"
+run_arm hooked2     "$HOOK"  "IGNORE="
+run_arm stock2      "$STOCK" "IGNORE="
+grep -E "rc=|loss" gpurun_out/ab5.log
+
+# Allocation-size census of the hooked run: which pool allocations
+# convert to managed (the churn suspects)?
+timeout 200 env NVSHARE_DEBUG=1 python -m nvshare_amd.run --standalone -- \
+  python -c "$TRAIN" > gpurun_out/ab5_dbg.log 2>&1
+grep -E "\+alloc" gpurun_out/ab5_dbg.log | \
+  awk '{print $4, $5, $NF}' | sort | uniq -c | sort -rn | head -30
+grep -c "passthrough" gpurun_out/ab5_dbg.log
