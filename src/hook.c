@@ -479,9 +479,100 @@ static size_t untrack_alloc2(void *ptr, int *was_passthrough)
 	return size;
 }
 
-static size_t untrack_alloc(void *ptr)
+/*
+ * Managed-range free cache.  Measured (profiles/ab5 census): the
+ * torch/MIOpen stack allocates and frees one small (2-128 MiB)
+ * segment per training step; every fresh hipMallocManaged range
+ * first-touch faults at page speed, which is exactly the episodic
+ * 100-215 ms kernel-call tail that costs the ~1.4x managed step
+ * overhead (profiles/RESULTS.md §14).  Freed managed ranges are
+ * therefore kept (pages stay resident) and recycled on the next
+ * same-size allocation.  Bounded by NVSHARE_FREE_CACHE_MIB (default
+ * 1024); flushed when the cap would reject an allocation and before
+ * pressure eviction.  Note hipFree's implicit device synchronization
+ * is skipped for cached frees — the same contract relaxation the
+ * managed conversion already makes.
+ */
+struct nvs_cached {
+	void *ptr;
+	size_t size;
+	struct nvs_cached *next;
+};
+static struct nvs_cached *free_cache;
+static size_t free_cache_bytes;
+
+static size_t free_cache_cap(void)
 {
-	return untrack_alloc2(ptr, NULL);
+	static long mib = -1;
+
+	if (mib < 0)
+		mib = nvs_env_long("NVSHARE_FREE_CACHE_MIB", 1024, 0,
+				   1024 * 1024);
+	return (size_t)mib * NVS_MIB;
+}
+
+/* Caller must NOT hold alloc_mutex. */
+static void flush_free_cache(void)
+{
+	struct nvs_cached *list, *c;
+
+	pthread_mutex_lock(&alloc_mutex);
+	list = free_cache;
+	free_cache = NULL;
+	free_cache_bytes = 0;
+	pthread_mutex_unlock(&alloc_mutex);
+	while (list != NULL) {
+		c = list;
+		list = c->next;
+		if (real.hipFree != NULL)
+			real.hipFree(c->ptr);
+		free(c);
+	}
+}
+
+/* Try to serve an allocation from the cache (exact size match; the
+ * per-step churn repeats identical sizes).  Returns ptr or NULL. */
+static void *free_cache_pop(size_t size)
+{
+	struct nvs_cached **pp, *c;
+	void *ptr = NULL;
+
+	pthread_mutex_lock(&alloc_mutex);
+	for (pp = &free_cache; *pp != NULL; pp = &(*pp)->next) {
+		if ((*pp)->size == size) {
+			c = *pp;
+			*pp = c->next;
+			ptr = c->ptr;
+			free_cache_bytes -= size;
+			free(c);
+			break;
+		}
+	}
+	pthread_mutex_unlock(&alloc_mutex);
+	return ptr;
+}
+
+/* Stash a managed range instead of freeing it.  Returns 1 when
+ * cached (caller must NOT call the real free). */
+static int free_cache_push(void *ptr, size_t size)
+{
+	struct nvs_cached *c;
+
+	if (size == 0 || size + free_cache_bytes > free_cache_cap())
+		return 0;
+	c = malloc(sizeof(*c));
+	if (c == NULL)
+		return 0;
+	c->ptr = ptr;
+	c->size = size;
+	pthread_mutex_lock(&alloc_mutex);
+	c->next = free_cache;
+	free_cache = c;
+	free_cache_bytes += size;
+	pthread_mutex_unlock(&alloc_mutex);
+	log_debug("hook: cached free %p %zu MiB (cache %zu MiB)", ptr,
+		  size / NVS_MIB, free_cache_bytes / NVS_MIB);
+	return 1;
 }
 
 /*
@@ -606,6 +697,9 @@ void nvs_evict_allocs(void)
 	size_t resident, freeb;
 	int64_t t0 = nvs_now_ns();
 
+	/* Held-but-freed ranges are dead weight under pressure: free
+	 * them outright instead of migrating them. */
+	flush_free_cache();
 	if (real.hipMemPrefetchAsync == NULL)
 		return;
 	ensure_prefetch_streams();
@@ -815,8 +909,24 @@ static nvshipError_t malloc_managed(void **ptr, size_t size)
 		return real.hipMalloc(ptr, size);
 	if (disable_um)
 		return real.hipMalloc(ptr, size);
-	if (!oversub_allowed && reserve_cap(size) != 0)
-		return NVSHIP_ERROR_OOM;
+	if (!oversub_allowed && reserve_cap(size) != 0) {
+		/* Cached (held-but-freed) ranges may be what's in the
+		 * way: release them and retry once. */
+		flush_free_cache();
+		if (reserve_cap(size) != 0)
+			return NVSHIP_ERROR_OOM;
+	}
+	{
+		void *cached = free_cache_pop(size);
+
+		if (cached != NULL) {
+			/* Recycled range: pages already resident and
+			 * advised — no first-touch fault storm. */
+			*ptr = cached;
+			track_alloc(cached, size, 1, !oversub_allowed);
+			return NVSHIP_SUCCESS;
+		}
+	}
 	r = real.hipMallocManaged(ptr, size, NVSHIP_MEM_ATTACH_GLOBAL);
 	if (r == NVSHIP_SUCCESS)
 		track_alloc(*ptr, size, populate_managed(*ptr, size),
@@ -940,8 +1050,14 @@ nvshipError_t hipFree(void *ptr)
 	BOOTSTRAP();
 	BUMP(H_hipFree);
 	CHECK_REAL(hipFree);
-	if (ptr != NULL)
-		untrack_alloc(ptr);
+	if (ptr != NULL) {
+		int was_pt = 0;
+		size_t sz = untrack_alloc2(ptr, &was_pt);
+
+		if (sz != 0 && !was_pt && !disable_um &&
+		    free_cache_push(ptr, sz))
+			return NVSHIP_SUCCESS;
+	}
 	return real.hipFree(ptr);
 }
 
@@ -953,13 +1069,21 @@ nvshipError_t hipFreeAsync(void *ptr, nvship_stream_t stream)
 	BUMP(H_hipFreeAsync);
 	if (ptr == NULL)
 		return NVSHIP_SUCCESS;
-	if (untrack_alloc2(ptr, &was_pt) != 0 && !was_pt) {
-		/* We converted this to a managed alloc: stream-ordered
-		 * free semantics require prior stream work to finish. */
-		CHECK_REAL(hipFree);
-		if (real.hipStreamSynchronize != NULL)
-			real.hipStreamSynchronize(stream);
-		return real.hipFree(ptr);
+	{
+		size_t sz = untrack_alloc2(ptr, &was_pt);
+
+		if (sz != 0 && !was_pt) {
+			/* We converted this to a managed alloc:
+			 * stream-ordered free semantics require prior
+			 * stream work to finish before the range can
+			 * be reused or released. */
+			CHECK_REAL(hipFree);
+			if (real.hipStreamSynchronize != NULL)
+				real.hipStreamSynchronize(stream);
+			if (!disable_um && free_cache_push(ptr, sz))
+				return NVSHIP_SUCCESS;
+			return real.hipFree(ptr);
+		}
 	}
 	CHECK_REAL(hipFreeAsync);
 	return real.hipFreeAsync(ptr, stream);

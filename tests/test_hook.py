@@ -332,6 +332,48 @@ def test_client_env_strips_expandable_segments():
     assert "PYTORCH_CUDA_ALLOC_CONF" not in env
 
 
+def test_managed_free_cache_recycles(artifacts, sched, sock_dir):
+    """Freed managed ranges are held and recycled on same-size
+    reallocation (one real hipMallocManaged for N alloc/free cycles)
+    — the fix for the per-step segment churn that caused the episodic
+    fault-storm overhead (profiles/RESULTS.md §14)."""
+    env = client_env(sock_dir=sock_dir, use_stub=True, debug=True,
+                     reserve_mib=64)
+    env["NVSTUB_TOTAL_MIB"] = "4096"
+    log = os.path.join(sock_dir, "cache.log")
+    env["NVSTUB_LOG"] = log
+    code = r"""
+#include <stdio.h>
+typedef int hipError_t;
+extern hipError_t hipMalloc(void **, unsigned long);
+extern hipError_t hipFree(void *);
+int main(void) {
+    void *p; int i;
+    for (i = 0; i < 5; i++) {
+        if (hipMalloc(&p, 64ul << 20) != 0) return 1;
+        if (hipFree(p) != 0) return 2;
+    }
+    puts("CACHE_OK");
+    return 0;
+}
+"""
+    src = os.path.join(sock_dir, "cache.c")
+    exe = os.path.join(sock_dir, "cachex")
+    with open(src, "w") as f:
+        f.write(code)
+    build = subprocess.run(
+        ["gcc", "-o", exe, src, "-L", str(artifacts.stub_dir),
+         "-lamdhip64", f"-Wl,-rpath,{artifacts.stub_dir}"],
+        capture_output=True, text=True)
+    assert build.returncode == 0, build.stderr
+    r = subprocess.run([exe], env=env, capture_output=True, text=True,
+                       timeout=60)
+    assert r.returncode == 0, (r.stdout, r.stderr)
+    names = [e[2] for e in load_events(log)]
+    assert names.count("hipMallocManaged") == 1, names
+    assert names.count("hipFree") == 0, names  # all frees cached
+
+
 def test_fork_safety(artifacts, sched, sock_dir):
     """A forked child free-runs without corrupting the parent's
     scheduler protocol or deadlocking (reference would deadlock)."""
