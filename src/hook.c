@@ -104,6 +104,12 @@ static int alloc_prefetch = 0;   /* NVSHARE_ALLOC_PREFETCH (see
 				  * 7.2, default off) */
 static int coarse_grain = 1;     /* NVSHARE_COARSE_GRAIN */
 static int preferred_loc = 1;    /* NVSHARE_PREFERRED_LOC */
+static int alloc_memset = 0;     /* NVSHARE_ALLOC_MEMSET: populate
+				  * fresh managed ranges with a
+				  * device-side zero fill at alloc
+				  * time (pages born in HBM; synced
+				  * before the pointer is returned,
+				  * so nothing can race it) */
 static int thp_advise = 0;       /* NVSHARE_THP: MADV_HUGEPAGE on
 				  * managed ranges (experiment knob:
 				  * 2 MiB mappings cut device TLB
@@ -232,6 +238,7 @@ static void bootstrap(void)
 	coarse_grain = nvs_env_bool("NVSHARE_COARSE_GRAIN", 1);
 	preferred_loc = nvs_env_bool("NVSHARE_PREFERRED_LOC", 1);
 	thp_advise = nvs_env_bool("NVSHARE_THP", 0);
+	alloc_memset = nvs_env_bool("NVSHARE_ALLOC_MEMSET", 0);
 	/* Reserve sized for 288 GB HBM3E; the reference used 1536 MiB on a
 	 * 16 GB P100 (hook.c:45). */
 	mem_reserve = (size_t)nvs_env_long("NVSHARE_RESERVE_MIB", 8192, 0,
@@ -331,6 +338,17 @@ static int populate_managed(void *ptr, size_t size)
 		real.hipMemAdvise(ptr, size,
 				  NVSHIP_MEM_ADVISE_SET_PREFERRED_LOCATION,
 				  nvs_app_device);
+	if (alloc_memset && real.hipMemset != NULL &&
+	    real.hipDeviceSynchronize != NULL) {
+		/* Device-side first touch: the fill kernel's writes
+		 * materialize the pages in HBM (preferred location),
+		 * and the sync completes before the app ever sees the
+		 * pointer — safe by construction, no prefetch
+		 * machinery involved. */
+		real.hipMemset(ptr, 0, size);
+		real.hipDeviceSynchronize();
+		return 1;
+	}
 	if (!alloc_prefetch || real.hipMemPrefetchAsync == NULL)
 		return 1; /* nothing to defer */
 	if (!nvs_can_submit_now())
