@@ -10,6 +10,34 @@ set -x
 mkdir -p gpurun_out
 export HSA_XNACK=1 MIOPEN_FIND_MODE=FAST
 
+# 0. XNACK-mode arm: is the steady ~1.35x managed tax the XNACK
+#    retry-fault servicing itself?  HSA_XNACK=0 managed memory uses
+#    legacy whole-buffer residency (no retry faults).
+TRAIN='
+import sys; sys.path.insert(0, ".")
+from nvshare_amd.workloads.train_resnet import run_training
+import math
+r = run_training("resnet50", "cuda", batch=32, image=224, steps=60,
+                 warmup=10)
+assert math.isfinite(r["loss"]), r
+print("ARM", round(r["samples_per_s"], 1), "loss", round(r["loss"], 3))
+'
+timeout 300 env -u LD_PRELOAD python -c "
+import sys; sys.path.insert(0, '.')
+from nvshare_amd.workloads.train_resnet import run_training
+run_training('resnet50', 'cuda', batch=32, image=224, steps=3, warmup=3)
+print('WARMED')" > gpurun_out/sc_warm.log 2>&1
+: > gpurun_out/xnack.log
+for arm in "stock_x1 env -u LD_PRELOAD" "hooked_x1 python -m nvshare_amd.run --standalone --"; do
+  set -- $arm; label=$1; shift
+  timeout 200 env HSA_XNACK=1 "$@" python -c "$TRAIN" 2>&1 | sed "s/^ARM/$label/" >> gpurun_out/xnack.log
+done
+for arm in "stock_x0 env -u LD_PRELOAD" "hooked_x0 python -m nvshare_amd.run --standalone --" "hooked_x0b python -m nvshare_amd.run --standalone --"; do
+  set -- $arm; label=$1; shift
+  timeout 200 env HSA_XNACK=0 "$@" python -c "$TRAIN" 2>&1 | sed "s/^ARM/$label/" >> gpurun_out/xnack.log
+done
+grep -E "^stock|^hooked" gpurun_out/xnack.log
+
 # 1. N-client curve (K scales with N so per-rank work stays constant).
 for N in 1 2 4 8; do
   timeout 420 python -m torch.distributed.run --nnodes=1 \
