@@ -36,7 +36,9 @@ def client_env(
     if env.get("LD_PRELOAD"):
         preload = preload + ":" + env["LD_PRELOAD"]
     env["LD_PRELOAD"] = preload
-    env["HSA_XNACK"] = "1"
+    # gfx950 page-granular demand paging; an explicit operator setting
+    # (e.g. HSA_XNACK=0 for legacy whole-buffer residency) wins.
+    env.setdefault("HSA_XNACK", "1")
 
     # PyTorch's expandable-segments allocator backend goes through
     # hipMemCreate/hipMemMap: the interposer caps it (hook.c) but the
