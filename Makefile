@@ -13,9 +13,10 @@ test: all
 
 tsan:
 	$(MAKE) -C src BUILD=build-tsan \
-	    CFLAGS="-O1 -g -Wall -Wextra -std=gnu11 -fPIC -fsanitize=thread" \
+	    CFLAGS="-O1 -g -Wall -Wextra -std=gnu11 -fPIC -fsanitize=thread -DNVSHARE_NO_DLSYM_EXPORT" \
 	    LDFLAGS_HARDEN="-fsanitize=thread" \
-	    build-tsan/nvshare-scheduler
+	    build-tsan/nvshare-scheduler build-tsan/libnvshare.so \
+	    build-tsan/libamdhip64.so.7 build-tsan/hipclient
 
 clean:
 	$(MAKE) -C src clean

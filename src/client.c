@@ -79,7 +79,12 @@ static pthread_rwlock_t submit_rwlock = PTHREAD_RWLOCK_INITIALIZER;
 static pthread_mutex_t sock_mutex = PTHREAD_MUTEX_INITIALIZER;
 static int sock_fd = -1;
 
-static volatile int shutting_down;
+static int shutting_down; /* atomic accesses only */
+
+static inline int is_shutting_down(void)
+{
+	return __atomic_load_n(&shutting_down, __ATOMIC_RELAXED);
+}
 
 static sem_t init_done_sem;
 static char pod_name[NVS_POD_NAME_LEN];
@@ -187,9 +192,9 @@ int nvs_can_submit_now(void)
 static void drain_gpu(void)
 {
 	pthread_rwlock_wrlock(&submit_rwlock);
-	if (!shutting_down && real.hipSetDevice != NULL)
+	if (!is_shutting_down() && real.hipSetDevice != NULL)
 		real.hipSetDevice(nvs_app_device);
-	if (!shutting_down && real.hipDeviceSynchronize != NULL)
+	if (!is_shutting_down() && real.hipDeviceSynchronize != NULL)
 		real.hipDeviceSynchronize();
 	pthread_rwlock_unlock(&submit_rwlock);
 }
@@ -199,7 +204,7 @@ static void drain_gpu(void)
  * with atexit (runs before library destructors). */
 static void on_process_exit(void)
 {
-	shutting_down = 1;
+	__atomic_store_n(&shutting_down, 1, __ATOMIC_RELAXED);
 }
 
 static void read_pod_identity(void)
@@ -428,7 +433,7 @@ static void idle_detect_init(void)
 /* 1 = GPU looks idle, 0 = busy/unknown. */
 static int check_idle(void)
 {
-	if (shutting_down)
+	if (is_shutting_down())
 		return 0;
 	if (rsmi_ready) {
 		uint32_t busy = 100;
@@ -465,7 +470,7 @@ static void *early_release_thread(void *arg)
 		long mem_mib;
 
 		usleep((useconds_t)(release_interval_ms * 1000));
-		if (shutting_down)
+		if (is_shutting_down())
 			continue; /* never touch HIP during teardown */
 		mem_mib = nvs_sum_allocated_mib();
 		if (mem_mib != last_mem_mib) {

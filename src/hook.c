@@ -1750,5 +1750,11 @@ void *nvs_dlsym_225(void *handle, const char *name)
 	return nvs_dlsym_common(handle, name);
 }
 
+/* TSan builds skip the dlsym@GLIBC export: ThreadSanitizer's own
+ * dlsym interceptor collides with it before its runtime initializes
+ * (build-tsan is for race checking the threading logic, not the
+ * loader tricks). */
+#ifndef NVSHARE_NO_DLSYM_EXPORT
 __asm__(".symver nvs_dlsym_234, dlsym@@GLIBC_2.34");
 __asm__(".symver nvs_dlsym_225, dlsym@GLIBC_2.2.5");
+#endif

@@ -74,3 +74,40 @@ def test_scheduler_lifecycle_under_tsan(tsan_bin, sock_dir):
     text = Path(sock_dir, "tsan.log").read_text(errors="replace")
     assert "WARNING: ThreadSanitizer" not in text, text[-4000:]
     assert rc != 66, "TSan reported races"
+
+
+TSAN_DIR = REPO / "src" / "build-tsan"
+
+
+def test_client_library_under_tsan(tsan_bin, sock_dir):
+    """The interposer + client runtime (gate rwlock, client thread,
+    early-release thread, allocation tracker, free cache) run a
+    two-client preemption workload under ThreadSanitizer."""
+    if not (TSAN_DIR / "libnvshare.so").exists():
+        pytest.skip("tsan libnvshare unavailable")
+    from nvshare_amd.scheduler import SchedulerDaemon
+
+    with SchedulerDaemon(sock_dir=sock_dir, tq=1, debug=False):
+        procs = []
+        for i in range(2):
+            env = dict(os.environ)
+            env["LD_PRELOAD"] = str(TSAN_DIR / "libnvshare.so")
+            env["LD_LIBRARY_PATH"] = str(TSAN_DIR)
+            env["NVSHARE_SOCK_DIR"] = sock_dir
+            env["NVSHARE_POD_NAME"] = f"tsan{i}"
+            env["NVSHARE_RELEASE_INTERVAL_MS"] = "100"
+            env["NVSHARE_RESERVE_MIB"] = "64"
+            env["NVSTUB_TOTAL_MIB"] = "1024"
+            env["HSA_XNACK"] = "1"
+            env["TSAN_OPTIONS"] = "exitcode=66 halt_on_error=0"
+            p = subprocess.Popen(
+                [str(TSAN_DIR / "hipclient"), "--allocs", "4",
+                 "--alloc-mib", "8", "--iters", "400", "--sleep-us",
+                 "2000", "--sync-every", "50"],
+                env=env, stdout=subprocess.PIPE,
+                stderr=subprocess.PIPE, text=True)
+            procs.append(p)
+        for p in procs:
+            out, err = p.communicate(timeout=120)
+            assert "WARNING: ThreadSanitizer" not in err, err[-4000:]
+            assert p.returncode == 0, (p.returncode, out, err[-2000:])
