@@ -1118,7 +1118,13 @@ nvshipError_t hipMemGetInfo(size_t *free_p, size_t *total_p)
 		return real.hipMemGetInfo(free_p, total_p);
 	limit = mem_limit();
 	pthread_mutex_lock(&alloc_mutex);
-	freeb = sum_allocated < limit ? limit - sum_allocated : 0;
+	{
+		/* Held-but-freed (cached) ranges still occupy memory:
+		 * every byte counted (reference hook.c:662-670). */
+		size_t used = sum_allocated + free_cache_bytes;
+
+		freeb = used < limit ? limit - used : 0;
+	}
 	pthread_mutex_unlock(&alloc_mutex);
 	if (free_p != NULL)
 		*free_p = freeb;
