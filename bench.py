@@ -110,7 +110,7 @@ def reexec_under_preload(args, device: str) -> None:
             # all ranks co-locate on physical GPU 0
             "HIP_VISIBLE_DEVICES": os.environ.get(
                 "NVSHARE_BENCH_GPU", "0"),
-            "NVSHARE_RELEASE_INTERVAL_MS": "50",
+            "NVSHARE_RELEASE_INTERVAL_MS": "20",
             "NVSHARE_POD_NAME": f"bench-rank{rank}",
         },
     )

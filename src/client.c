@@ -435,7 +435,12 @@ static int check_idle(void)
 {
 	if (is_shutting_down())
 		return 0;
-	if (rsmi_ready) {
+	/* rsmi busy%% is windowed and lags work completion by hundreds
+	 * of ms — fine for the reference's 5 s cadence, but it IS the
+	 * handoff latency when probing fast.  Sub-second cadences use
+	 * the precise timed-sync probe instead (an idle device answers
+	 * hipDeviceSynchronize in microseconds). */
+	if (rsmi_ready && release_interval_ms >= 1000) {
 		uint32_t busy = 100;
 
 		if (p_rsmi_busy(rsmi_dev_index, &busy) == 0)
