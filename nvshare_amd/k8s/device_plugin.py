@@ -129,6 +129,11 @@ class DevicePluginServicer:
                 self.cfg.lib_dir, "libnvshare.so")
             cresp.envs["HSA_XNACK"] = "1"
             cresp.envs["NVSHARE_SOCK_DIR"] = self.cfg.sock_dir + "/"
+            # The node-level GPU index: inside the container the GPU
+            # is always HIP device 0, so the client needs this to
+            # register on the right per-GPU arbitration queue
+            # (src/client.c detect_physical_gpu).
+            cresp.envs["NVSHARE_GPU"] = str(gpu)
             if self.cfg.debug_clients:
                 cresp.envs["NVSHARE_DEBUG"] = "1"
 

@@ -90,6 +90,9 @@ def test_allocate_injects_env_and_mounts(plugin):
     assert c.envs["LD_PRELOAD"] == "/usr/lib/nvshare/libnvshare.so"
     assert c.envs["HSA_XNACK"] == "1"
     assert c.envs["NVSHARE_SOCK_DIR"] == "/var/run/nvshare/"
+    # Containers see their GPU as HIP device 0; the node-level index
+    # routes the client to the right per-GPU arbitration queue.
+    assert c.envs["NVSHARE_GPU"] == "0"
     paths = [d.container_path for d in c.devices]
     assert "/dev/kfd" in paths
     mounts = {m.container_path: m for m in c.mounts}
@@ -263,3 +266,47 @@ def test_preferred_allocation(plugin):
         resp = rpc(req, timeout=5)
     assert list(resp.container_responses[0].deviceIDs) == \
         ["gpu0__0", "gpu0__1"]
+
+
+def test_allocate_env_drives_client_to_gpu_queue(plugin, sock_dir,
+                                                 artifacts):
+    """End-to-end bridge: the EXACT env the plugin injects (plus the
+    CPU stub lib path) makes a client register on the allocated GPU's
+    queue of a live scheduler."""
+    import os
+    import subprocess
+
+    from nvshare_amd import ctl
+    from nvshare_amd.scheduler import SchedulerDaemon
+
+    req = pb.AllocateRequest()
+    creq = req.container_requests.add()
+    creq.devicesIDs.append("gpu3__1")
+    with plugin_channel(plugin) as chan:
+        rpc = chan.unary_unary(
+            f"/{pb.DEVICE_PLUGIN_SERVICE}/Allocate",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=pb.AllocateResponse.FromString)
+        resp = rpc(req, timeout=5)
+    c = resp.container_responses[0]
+    assert c.envs["NVSHARE_GPU"] == "3"
+
+    with SchedulerDaemon(sock_dir=sock_dir, tq=5, debug=True) as sched:
+        env = dict(os.environ)
+        env.update(dict(c.envs))
+        # Container-equivalent substitutions for the CPU harness: the
+        # real mounts map libnvshare.so + the socket dir into the pod.
+        env["LD_PRELOAD"] = str(artifacts.libnvshare)
+        env["NVSHARE_SOCK_DIR"] = sock_dir
+        env["LD_LIBRARY_PATH"] = str(artifacts.stub_dir)
+        env["NVSTUB_TOTAL_MIB"] = "1024"
+        env["NVSHARE_RESERVE_MIB"] = "64"
+        env["NVSHARE_DEBUG"] = "1"
+        r = subprocess.run(
+            [str(artifacts.hipclient), "--allocs", "1", "--alloc-mib",
+             "8", "--iters", "5"],
+            env=env, capture_output=True, text=True, timeout=60)
+        assert r.returncode == 0, (r.stdout, r.stderr[-2000:])
+        assert "PASS" in r.stdout
+        log = sched.log_text()
+        assert "on gpu3" in log, log[-2000:]
