@@ -598,11 +598,11 @@ static void dump_state(void)
 		if (!g->grants && !queue_len(g))
 			continue;
 		log_info(" gpu%d: lock_held=%d holder=%016" PRIx64
-			 " round=%lu grants=%lu preempts=%lu queue=%d", i,
-			 g->lock_held,
+			 " round=%lu grants=%lu preempts=%lu queue=%d "
+			 "pressure=%d", i, g->lock_held,
 			 g->lock_holder ? g->lock_holder->id : 0,
 			 g->round, g->grants, g->preemptions,
-			 queue_len(g));
+			 queue_len(g), gpu_pressure(g));
 	}
 	for (c = clients; c != NULL; c = c->next)
 		log_info("  conn fd=%d reg=%d id=%016" PRIx64 " gpu=%d "
