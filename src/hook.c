@@ -576,7 +576,7 @@ static int free_cache_push(void *ptr, size_t size)
 {
 	struct nvs_cached *c;
 
-	if (size == 0 || size + free_cache_bytes > free_cache_cap())
+	if (size == 0 || size > free_cache_cap())
 		return 0;
 	c = malloc(sizeof(*c));
 	if (c == NULL)
@@ -584,6 +584,11 @@ static int free_cache_push(void *ptr, size_t size)
 	c->ptr = ptr;
 	c->size = size;
 	pthread_mutex_lock(&alloc_mutex);
+	if (size + free_cache_bytes > free_cache_cap()) {
+		pthread_mutex_unlock(&alloc_mutex);
+		free(c);
+		return 0;
+	}
 	c->next = free_cache;
 	free_cache = c;
 	free_cache_bytes += size;
@@ -1168,6 +1173,7 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 {
 	static int enabled = -1;
 	static long min_mib = 8;
+	static pthread_mutex_t prep_mutex = PTHREAD_MUTEX_INITIALIZER;
 	struct nvs_alloc *a;
 	char *base = NULL, *d = dst;
 	size_t size = 0;
@@ -1189,6 +1195,10 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 		return;
 	if (n < (size_t)min_mib * NVS_MIB)
 		return;
+	/* Serialized end to end: a second large writer into the same
+	 * range must wait here until the first writer's migration has
+	 * fully completed, not just until the flag flips. */
+	pthread_mutex_lock(&prep_mutex);
 	pthread_mutex_lock(&alloc_mutex);
 	for (a = alloc_list; a != NULL; a = a->next) {
 		char *p = a->ptr;
@@ -1203,8 +1213,10 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 		}
 	}
 	pthread_mutex_unlock(&alloc_mutex);
-	if (base == NULL)
+	if (base == NULL) {
+		pthread_mutex_unlock(&prep_mutex);
 		return;
+	}
 	/* Quiesce the device first: an in-flight write into the same
 	 * range from ANOTHER stream would race the migration (the
 	 * documented-unsafe overlap).  Once per range, so the full
@@ -1214,6 +1226,7 @@ static void copy_prefetch(void *dst, size_t n, nvship_stream_t s)
 	real.hipMemPrefetchAsync(base, size, nvs_app_device, s);
 	real.hipStreamSynchronize(s); /* migration MUST complete before
 				       * the write lands (see above) */
+	pthread_mutex_unlock(&prep_mutex);
 }
 
 #define GATED2(id, call)                                                   \
