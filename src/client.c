@@ -550,6 +550,7 @@ static void atfork_child(void)
 	scheduler_on = 0;
 	own_lock = 0;
 	need_lock = 0;
+	nvs_free_cache_forget();
 }
 
 void nvs_client_init(void)

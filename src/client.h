@@ -103,4 +103,7 @@ long nvs_sum_allocated_mib(void);
 /* Advertised device capacity in MiB (hook.c). */
 long nvs_mem_total_mib(void);
 
+/* Drop the managed free cache without releasing (fork child). */
+void nvs_free_cache_forget(void);
+
 #endif /* NVSHARE_CLIENT_H */

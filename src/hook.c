@@ -529,6 +529,21 @@ static size_t free_cache_cap(void)
 	return (size_t)mib * NVS_MIB;
 }
 
+/* fork() child: drop the cache VIEW without freeing — the ranges
+ * belong to the parent's GPU context; recycling them in the child
+ * would alias live parent memory (client.c atfork_child). */
+void nvs_free_cache_forget(void)
+{
+	struct nvs_cached *c;
+
+	while (free_cache != NULL) {
+		c = free_cache;
+		free_cache = c->next;
+		free(c);
+	}
+	free_cache_bytes = 0;
+}
+
 /* Caller must NOT hold alloc_mutex. */
 static void flush_free_cache(void)
 {
