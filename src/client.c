@@ -58,6 +58,19 @@ static uint64_t client_id = 0;
  * NVSHARE_PREFETCH/NVSHARE_EVICT are not explicitly set. */
 static int sched_pressure = -1;
 
+/* Sharing metrics (atomic; dumped at exit when NVSHARE_DEBUG):
+ * - wait_ns: app-thread time blocked at the gate waiting for a grant
+ * - held_ns: lock tenure (grant .. release)
+ * - grants/preempts/early_releases: lifecycle counters */
+static int64_t metric_wait_ns;
+static int64_t metric_held_ns;
+static int64_t metric_grants;
+static int64_t metric_preempts;
+static int64_t metric_early_releases;
+static int64_t grant_t0_ns; /* guarded by g_mutex */
+
+#define METRIC_ADD(var, v) __atomic_fetch_add(&(var), (v), 					      __ATOMIC_RELAXED)
+
 /* Migration-assist policy: an explicit env setting wins; else the
  * scheduler's pressure verdict; else the AUTO_MIGRATE default (which
  * self-gates on real free memory in hook.c). */
@@ -145,6 +158,9 @@ void nvs_submit_begin(void)
 			return; /* read lock held */
 		}
 		pthread_rwlock_unlock(&submit_rwlock);
+		{
+			int64_t w0 = nvs_now_ns();
+
 		for (;;) {
 			struct timespec abs;
 
@@ -166,6 +182,8 @@ void nvs_submit_begin(void)
 						   &abs) != 0 &&
 			    scheduler_on && !own_lock)
 				need_lock = 0;
+		}
+			METRIC_ADD(metric_wait_ns, nvs_now_ns() - w0);
 		}
 		pthread_mutex_unlock(&g_mutex);
 	}
@@ -205,6 +223,14 @@ static void drain_gpu(void)
 static void on_process_exit(void)
 {
 	__atomic_store_n(&shutting_down, 1, __ATOMIC_RELAXED);
+	if (nvs_debug_enabled && !standalone)
+		log_debug("client sharing metrics: grants=%lld "
+			  "preempts=%lld early_releases=%lld "
+			  "held=%.1fs waited=%.1fs",
+			  (long long)metric_grants,
+			  (long long)metric_preempts,
+			  (long long)metric_early_releases,
+			  metric_held_ns / 1e9, metric_wait_ns / 1e9);
 }
 
 static void read_pod_identity(void)
@@ -300,6 +326,8 @@ static void handle_lock_ok(const struct nvs_msg *m)
 	pthread_mutex_lock(&g_mutex);
 	own_lock = 1;
 	need_lock = 0;
+	grant_t0_ns = nvs_now_ns();
+	METRIC_ADD(metric_grants, 1);
 	pthread_cond_broadcast(&own_lock_cv);
 	pthread_mutex_unlock(&g_mutex);
 }
@@ -311,6 +339,10 @@ static void handle_drop_lock(void)
 	pthread_mutex_lock(&g_mutex);
 	had = own_lock;
 	own_lock = 0;
+	if (had) {
+		METRIC_ADD(metric_held_ns, nvs_now_ns() - grant_t0_ns);
+		METRIC_ADD(metric_preempts, 1);
+	}
 	pthread_mutex_unlock(&g_mutex);
 	if (!had)
 		return; /* already released voluntarily */
@@ -510,6 +542,8 @@ static void *early_release_thread(void *arg)
 			continue;
 		}
 		own_lock = 0;
+		METRIC_ADD(metric_held_ns, nvs_now_ns() - grant_t0_ns);
+		METRIC_ADD(metric_early_releases, 1);
 		pthread_mutex_unlock(&g_mutex);
 		drain_gpu();
 		/* Same pressure-gated eviction as the DROP_LOCK path:

@@ -447,3 +447,33 @@ def test_scheduler_restart_reconnect(artifacts, sock_dir):
         assert "reconnecting" in err
     finally:
         d2.stop()
+
+
+def test_client_metrics_dump(artifacts, sched, sock_dir):
+    """NVSHARE_DEBUG exit dump reports sharing metrics (grants,
+    preemptions, held/waited time) — the operator-facing breakdown of
+    sharing overhead."""
+    import threading
+
+    results = []
+
+    def one(i):
+        r = run_hipclient(artifacts, sock_dir, "--allocs", 1,
+                          "--alloc-mib", 8, "--iters", 300,
+                          "--sleep-us", "5000", "--sync-every", 50,
+                          stub_env={"NVSTUB_TOTAL_MIB": "1024"},
+                          reserve_mib=64, timeout=60)
+        results.append(r)
+
+    ts = [threading.Thread(target=one, args=(i,)) for i in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    for r in results:
+        assert r.returncode == 0, r.stderr[-1500:]
+        assert "client sharing metrics:" in r.stderr
+        line = [l for l in r.stderr.splitlines()
+                if "sharing metrics" in l][0]
+        grants = int(line.split("grants=")[1].split()[0])
+        assert grants >= 1, line
