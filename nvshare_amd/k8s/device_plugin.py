@@ -81,6 +81,13 @@ class PluginConfig:
                                         RESOURCE_NAME)
     gpus: list[int] = field(default_factory=discover_gpus)
     debug_clients: bool = bool(os.environ.get("NVSHARE_CLIENT_DEBUG"))
+    # "managed": hipMalloc -> hipMallocManaged (oversubscribable, the
+    # reference behavior; costs ~1.4x on conv-heavy training, ROCm 7.2
+    # XNACK servicing — profiles/RESULTS.md §18).
+    # "gate-only": real VRAM + scheduler arbitration + allocation cap
+    # only (zero measured overhead; no oversubscription) — for fleets
+    # whose co-located working sets always fit in 288 GB.
+    client_mode: str = os.environ.get("NVSHARE_CLIENT_MODE", "managed")
 
 
 class DevicePluginServicer:
@@ -134,6 +141,8 @@ class DevicePluginServicer:
             # register on the right per-GPU arbitration queue
             # (src/client.c detect_physical_gpu).
             cresp.envs["NVSHARE_GPU"] = str(gpu)
+            if self.cfg.client_mode == "gate-only":
+                cresp.envs["NVSHARE_DISABLE_UM"] = "1"
             if self.cfg.debug_clients:
                 cresp.envs["NVSHARE_DEBUG"] = "1"
 

@@ -310,3 +310,28 @@ def test_allocate_env_drives_client_to_gpu_queue(plugin, sock_dir,
         assert "PASS" in r.stdout
         log = sched.log_text()
         assert "on gpu3" in log, log[-2000:]
+
+
+def test_gate_only_client_mode(kubelet_dir):
+    """NVSHARE_CLIENT_MODE=gate-only deployments inject
+    NVSHARE_DISABLE_UM=1: real-VRAM clients, scheduler arbitration and
+    cap only (zero-overhead sharing for fits-in-HBM fleets,
+    profiles/RESULTS.md §18)."""
+    cfg = PluginConfig(virtual_devices=2, kubelet_dir=kubelet_dir,
+                       gpus=[0], client_mode="gate-only")
+    p = DevicePlugin(cfg)
+    p.start()
+    try:
+        req = pb.AllocateRequest()
+        creq = req.container_requests.add()
+        creq.devicesIDs.append("gpu0__0")
+        with plugin_channel(p) as chan:
+            rpc = chan.unary_unary(
+                f"/{pb.DEVICE_PLUGIN_SERVICE}/Allocate",
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=pb.AllocateResponse.FromString)
+            resp = rpc(req, timeout=5)
+        c = resp.container_responses[0]
+        assert c.envs["NVSHARE_DISABLE_UM"] == "1"
+    finally:
+        p.stop()
