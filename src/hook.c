@@ -97,6 +97,9 @@ static pthread_mutex_t alloc_mutex = PTHREAD_MUTEX_INITIALIZER;
 
 static size_t mem_total;        /* advertised total (bytes) */
 static size_t mem_reserve;      /* carve-out (bytes) */
+static size_t free_cache_bytes; /* held-but-freed cache (guarded by
+				 * alloc_mutex; see the free-cache
+				 * block below) */
 static int oversub_allowed;
 static int disable_um;
 static int alloc_prefetch = 0;   /* NVSHARE_ALLOC_PREFETCH (see
@@ -398,7 +401,10 @@ static void track_alloc(void *ptr, size_t size, int populated,
 static int reserve_cap(size_t size)
 {
 	pthread_mutex_lock(&alloc_mutex);
-	if (sum_allocated + size > mem_limit()) {
+	/* Held-but-freed (cached) ranges still occupy memory and count
+	 * here too; the caller flushes the cache and retries when this
+	 * rejects (malloc_managed). */
+	if (sum_allocated + free_cache_bytes + size > mem_limit()) {
 		size_t sum = sum_allocated;
 
 		pthread_mutex_unlock(&alloc_mutex);
@@ -435,8 +441,8 @@ long nvs_sum_allocated_mib(void)
  * can judge whole-node memory pressure. */
 long nvs_mem_total_mib(void)
 {
-	return (long)(mem_limit() / NVS_MIB) +
-	       (long)(mem_reserve / NVS_MIB);
+	(void)mem_limit(); /* ensure memquery ran */
+	return (long)(mem_total / NVS_MIB);
 }
 
 /* Called from the gate with the submission read lock held and the GPU
@@ -517,7 +523,6 @@ struct nvs_cached {
 	struct nvs_cached *next;
 };
 static struct nvs_cached *free_cache;
-static size_t free_cache_bytes;
 
 static size_t free_cache_cap(void)
 {
