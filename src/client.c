@@ -161,28 +161,29 @@ void nvs_submit_begin(void)
 		{
 			int64_t w0 = nvs_now_ns();
 
-		for (;;) {
-			struct timespec abs;
+			for (;;) {
+				struct timespec abs;
 
-			if (!scheduler_on || own_lock)
-				break;
-			if (!need_lock) {
-				need_lock = 1;
-				/* sock_mutex nests under g_mutex (never the
-				 * other way around). */
-				if (send_msg_type(NVS_REQ_LOCK) != 0)
-					need_lock = 0; /* retry next pass */
+				if (!scheduler_on || own_lock)
+					break;
+				if (!need_lock) {
+					need_lock = 1;
+					/* sock_mutex nests under g_mutex
+					 * (never the other way around). */
+					if (send_msg_type(NVS_REQ_LOCK) != 0)
+						need_lock = 0; /* retry */
+				}
+				/* Re-send REQ_LOCK if nothing arrives for
+				 * 10 s: idempotent at the scheduler, and
+				 * insurance against lost-wakeup bugs. */
+				clock_gettime(CLOCK_REALTIME, &abs);
+				abs.tv_sec += 10;
+				if (pthread_cond_timedwait(&own_lock_cv,
+							   &g_mutex,
+							   &abs) != 0 &&
+				    scheduler_on && !own_lock)
+					need_lock = 0;
 			}
-			/* Re-send REQ_LOCK if nothing arrives for 10 s:
-			 * idempotent at the scheduler, and insurance
-			 * against lost-wakeup protocol bugs. */
-			clock_gettime(CLOCK_REALTIME, &abs);
-			abs.tv_sec += 10;
-			if (pthread_cond_timedwait(&own_lock_cv, &g_mutex,
-						   &abs) != 0 &&
-			    scheduler_on && !own_lock)
-				need_lock = 0;
-		}
 			METRIC_ADD(metric_wait_ns, nvs_now_ns() - w0);
 		}
 		pthread_mutex_unlock(&g_mutex);
