@@ -533,3 +533,45 @@ def test_two_clients_survive_scheduler_restart(artifacts, sock_dir):
                                                  len(intervals))
     finally:
         d2.stop()
+
+
+def test_sched_off_clients_overlap(artifacts, sock_dir):
+    """With scheduling disabled the gate opens: co-located clients'
+    kernel windows DO overlap (free-for-all, reference README:341-354
+    semantics) — the sanity inverse of test_two_clients_serialized."""
+    from nvshare_amd.scheduler import SchedulerDaemon
+
+    log = os.path.join(sock_dir, "ev5.log")
+    stub_env = {"NVSTUB_LOG": log, "NVSTUB_TOTAL_MIB": "1024",
+                "NVSTUB_KERNEL_US": "2500"}
+    with SchedulerDaemon(sock_dir=sock_dir, tq=1, sched_off=True):
+        env = client_env(sock_dir=sock_dir, use_stub=True,
+                         reserve_mib=64)
+        env.update(stub_env)
+        procs = []
+        for i in range(2):
+            e = dict(env)
+            e["NVSHARE_POD_NAME"] = f"ff{i}"
+            procs.append(subprocess.Popen(
+                [str(artifacts.hipclient), "--allocs", "1",
+                 "--alloc-mib", "16", "--iters", "400"],
+                env=e, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                text=True))
+        for p in procs:
+            out, err = p.communicate(timeout=120)
+            assert p.returncode == 0, (out, err)
+            assert "PASS" in out
+
+    events = load_events(log)
+    intervals = []
+    open_t = {}
+    for ts, pid, name, arg in events:
+        if name == "launch_begin":
+            open_t[pid] = ts
+        elif name == "launch_end":
+            intervals.append((open_t[pid], ts, pid))
+    intervals.sort()
+    overlaps = sum(
+        1 for a, b in zip(intervals, intervals[1:])
+        if b[0] < a[1] and a[2] != b[2])
+    assert overlaps > 0, "free-for-all mode never overlapped"
