@@ -348,6 +348,20 @@ static void handle_drop_lock(void)
 	if (!had)
 		return; /* already released voluntarily */
 	drain_gpu();
+	/* NVSHARE_EVICT_ASYNC=1 (experiment, default off): release the
+	 * lock BEFORE evicting so the next holder's restore overlaps
+	 * our eviction — cuts the handoff critical path from
+	 * evict+restore to max(evict, restore) IF the two bulk
+	 * migrations don't contend (unvalidated on hardware; the
+	 * round-1 livelock was demand faults racing bulk migration,
+	 * a different pair). */
+	if (nvs_env_bool("NVSHARE_EVICT_ASYNC", 0)) {
+		send_msg_type(NVS_LOCK_RELEASED);
+		if (migrate_enabled("NVSHARE_EVICT"))
+			nvs_evict_allocs();
+		log_debug("client: lock released (async evict)");
+		return;
+	}
 	/* Eviction additionally self-gates on real memory pressure (it
 	 * is skipped when free HBM already fits the tracked set), so the
 	 * automatic default is safe for fitting workloads. */

@@ -575,3 +575,36 @@ def test_sched_off_clients_overlap(artifacts, sock_dir):
         1 for a, b in zip(intervals, intervals[1:])
         if b[0] < a[1] and a[2] != b[2])
     assert overlaps > 0, "free-for-all mode never overlapped"
+
+
+def test_evict_async_handoff(artifacts, sched, sock_dir):
+    """NVSHARE_EVICT_ASYNC=1 releases the lock before evicting (the
+    next holder's restore overlaps the eviction); both over-capacity
+    clients still finish and evictions still happen."""
+    import threading
+
+    results = []
+
+    def one(idx):
+        log = os.path.join(sock_dir, f"ae{idx}.log")
+        r = run_hipclient(
+            artifacts, sock_dir, "--allocs", 4, "--alloc-mib", 100,
+            "--iters", 400, "--sleep-us", 10000, "--sync-every", 50,
+            stub_env={"NVSTUB_LOG": log, "NVSTUB_TOTAL_MIB": "600"},
+            env_extra={"NVSHARE_RELEASE_INTERVAL_MS": "200",
+                       "NVSHARE_EVICT_ASYNC": "1"},
+            reserve_mib=64, timeout=120)
+        results.append((r, log))
+
+    ts = [threading.Thread(target=one, args=(i,)) for i in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    evicts = 0
+    for r, log in results:
+        assert r.returncode == 0, (r.stdout, r.stderr[-2000:])
+        assert "async evict" in r.stderr, r.stderr[-1500:]
+        names = [e[2] for e in load_events(log)]
+        evicts += names.count("hipMemPrefetchAsync_cpu")
+    assert evicts > 0
